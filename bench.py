@@ -1,0 +1,272 @@
+"""Flagship benchmark: MNIST-CNN Stage samples/sec on MI355X.
+
+Measures the BASELINE.json headline metric — samples/sec (whole node) of
+the MNIST-CNN training Stage — plus the secondary configs (ResNet-50
+bf16, GPT-2 small). Synthetic data (no network access), random-init
+weights.
+
+Launch (driver contract):
+    python bench.py --gpus 1 --steps K --warmup W
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+The timed region is exactly K full training steps (zero_grad, forward,
+loss, backward, gradient sync, optimizer step, metric tracking),
+bracketed by barrier + torch.cuda.synchronize on both sides; the
+reported time is the MAX over ranks.
+
+Fast path (default, --impl flat): the model is a FlatReplica (one flat
+fp32 parameter/grad buffer), the optimizer a fused gfx950 Adam kernel,
+gradient sync a single RCCL all-reduce, and the whole step is captured
+into a hipGraph and replayed — per-step host work is one small D2D batch
+copy plus one graph launch. --impl ddp runs the torch-DDP eager path for
+comparison.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+from dmlcloud_amd import TrainingPipeline, TrainValStage
+from dmlcloud_amd.metrics import Reduction
+from dmlcloud_amd.models import gpt2_small, gpt2_tiny, mnist_cnn, resnet50
+from dmlcloud_amd.parallel import (
+    FlatAdam,
+    FlatSGD,
+    GraphedStep,
+    init_process_group_auto,
+    local_rank,
+)
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=200)
+    p.add_argument('--warmup', type=int, default=30)
+    p.add_argument('--model', choices=['mnist', 'resnet50', 'gpt2'], default='mnist')
+    p.add_argument('--impl', choices=['flat', 'ddp'], default='flat')
+    p.add_argument('--batch-size', type=int, default=None, help='per-GPU batch size')
+    p.add_argument('--no-graph', action='store_true', help='disable hipGraph capture')
+    p.add_argument('--seq-len', type=int, default=1024, help='gpt2 sequence length')
+    return p.parse_args()
+
+
+class BenchStage(TrainValStage):
+    """TrainValStage driven step-by-step by the benchmark loop."""
+
+    def __init__(self, args, device):
+        super().__init__()
+        self.args = args
+        self.bench_device = device
+        self.metric_prefix = self.train_metric_prefix()
+        self.loss_fn = torch.nn.CrossEntropyLoss()
+        self.static_batch = None
+
+    # -- model-specific setup -------------------------------------------------
+
+    def pre_stage(self):
+        args = self.args
+        device = self.bench_device
+        torch.manual_seed(1234)
+
+        if args.model == 'mnist':
+            model = mnist_cnn()
+            self.batch_shape = (args.batch_size, 1, 28, 28)
+            self.dtype = 'fp32'
+        elif args.model == 'resnet50':
+            model = resnet50()
+            self.batch_shape = (args.batch_size, 3, 224, 224)
+            self.dtype = 'bf16'
+        else:  # gpt2
+            model = gpt2_small() if device.type == 'cuda' else gpt2_tiny()
+            self.batch_shape = (args.batch_size, args.seq_len)
+            self.dtype = 'bf16'
+
+        if args.impl == 'flat':
+            self.pipeline.register_model('net', model, ddp_impl='flat', verbose=False)
+            replica = self.pipeline.models['net']
+            if args.model == 'mnist':
+                self.pipeline.register_optimizer('opt', FlatAdam(replica, lr=1e-3))
+            else:
+                self.pipeline.register_optimizer('opt', FlatSGD(replica, lr=1e-3, momentum=0.9))
+        else:
+            use_ddp = dist.is_initialized() and dist.get_world_size() > 1
+            self.pipeline.register_model('net', model, use_ddp=use_ddp, verbose=False)
+            wrapped = self.pipeline.models['net']
+            self.pipeline.register_optimizer('opt', torch.optim.Adam(wrapped.parameters(), lr=1e-3))
+
+        # synthetic data, resident on device: a pool of distinct batches
+        g = torch.Generator(device='cpu').manual_seed(4242)
+        self.n_pool = 16
+        if args.model == 'gpt2':
+            vocab = self.pipeline.models['net'].module.cfg.vocab_size if hasattr(
+                self.pipeline.models['net'], 'module'
+            ) else model.cfg.vocab_size
+            self.pool = [
+                torch.randint(0, vocab, self.batch_shape, generator=g).to(device) for _ in range(self.n_pool)
+            ]
+            self.labels = None
+            self.static_batch = torch.zeros_like(self.pool[0])
+        else:
+            self.pool = [torch.randn(self.batch_shape, generator=g).to(device) for _ in range(self.n_pool)]
+            self.labels = [
+                torch.randint(0, 10 if args.model == 'mnist' else 1000, (args.batch_size,), generator=g).to(device)
+                for _ in range(self.n_pool)
+            ]
+            self.static_batch = torch.zeros_like(self.pool[0])
+            self.static_labels = torch.zeros_like(self.labels[0])
+
+    # -- the training step ----------------------------------------------------
+
+    def step(self, batch):
+        model = self.pipeline.models['net']
+        if self.args.model == 'gpt2':
+            idx = batch
+            with torch.autocast('cuda', dtype=torch.bfloat16, enabled=self.bench_device.type == 'cuda'):
+                _, loss = model(idx, targets=idx)
+            return loss
+        x, y = batch
+        if self.args.model == 'resnet50':
+            with torch.autocast('cuda', dtype=torch.bfloat16, enabled=self.bench_device.type == 'cuda'):
+                out = model(x)
+                loss = self.loss_fn(out, y)
+            return loss
+        out = model(x)  # mnist: fp32, matching the reference's precision
+        return self.loss_fn(out, y)
+
+    def core_step(self):
+        """The graph-capturable part of train_batch: full compute + loss
+        metric accumulation on static buffers."""
+        self.zero_grad()
+        if self.args.model == 'gpt2':
+            loss = self.train_step(self.static_batch)
+        else:
+            loss = self.train_step((self.static_batch, self.static_labels))
+        self.optimize(loss)
+        self.track_reduce(self.loss_metric_name(), loss)
+
+    def load_batch(self, i):
+        """Stage pool batch i into the static buffers (device-side copy)."""
+        j = i % self.n_pool
+        self.static_batch.copy_(self.pool[j], non_blocking=True)
+        if self.labels is not None:
+            self.static_labels.copy_(self.labels[j], non_blocking=True)
+
+
+def main():
+    args = parse_args()
+
+    if 'MASTER_ADDR' not in os.environ and args.gpus == 1:
+        pass  # dummy single-process group below
+    init_process_group_auto()
+    world = dist.get_world_size()
+    rank = dist.get_rank()
+
+    if torch.cuda.is_available():
+        lr = local_rank() if local_rank() is not None else 0
+        device = torch.device('cuda', lr)
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device('cpu')
+
+    if args.batch_size is None:
+        args.batch_size = {'mnist': 1024, 'resnet50': 256, 'gpt2': 8}[args.model]
+
+    # Stage + pipeline machinery (the metric path under test runs per step)
+    pipeline = TrainingPipeline(name='bench')
+    stage = BenchStage(args, device)
+    pipeline.append_stage(stage, max_epochs=1)
+    pipeline.device = device
+    stage.pre_stage()
+
+    use_graph = (not args.no_graph) and args.impl == 'flat' and device.type == 'cuda'
+
+    if use_graph:
+        stage.load_batch(0)
+        graphed = GraphedStep(stage.core_step, warmup=3)
+        graphed.initialize()
+
+        def run_step(i):
+            stage.load_batch(i)
+            graphed()
+
+    else:
+
+        def run_step(i):
+            stage.load_batch(i)
+            stage.core_step()
+
+    # ---- warmup ----
+    for i in range(args.warmup):
+        run_step(i)
+
+    if device.type == 'cuda':
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    if device.type == 'cuda':
+        torch.cuda.synchronize()
+
+    # ---- timed region: exactly K steps ----
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        run_step(args.warmup + i)
+    if device.type == 'cuda':
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+
+    if world > 1:
+        et = torch.tensor([elapsed], dtype=torch.float64, device=device if device.type == 'cuda' else 'cpu')
+        dist.all_reduce(et, op=dist.ReduceOp.MAX)
+        elapsed = et.item()
+        dist.barrier()
+    if device.type == 'cuda':
+        torch.cuda.synchronize()
+
+    samples = args.steps * args.batch_size * world
+    samples_per_sec = samples / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+
+    if args.model == 'gpt2':
+        metric, value = 'tokens/s', samples_per_sec * args.seq_len
+    else:
+        metric, value = 'samples/s', samples_per_sec
+
+    if rank == 0:
+        result = {
+            'metric': metric,
+            'value': value,
+            'unit': metric,
+            'n_gpus': world,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': ms_per_step,
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,  # reference publishes no numbers
+            'dtype': stage.dtype,
+            'data': 'synthetic',
+            'config': {
+                'model': {'mnist': 'mnist-cnn', 'resnet50': 'resnet-50', 'gpt2': 'gpt2-small'}[args.model],
+                'global_batch': args.batch_size * world,
+                'per_gpu_batch': args.batch_size,
+                'parallelism': f'dp{world}',
+                'impl': args.impl,
+                'hipgraph': bool(use_graph and getattr(graphed, 'captured', False)) if use_graph else False,
+                **({'seq_len': args.seq_len} if args.model == 'gpt2' else {}),
+            },
+        }
+        print(json.dumps(result))
+
+    dist.destroy_process_group()
+
+
+if __name__ == '__main__':
+    sys.exit(main() or 0)

@@ -1,0 +1,44 @@
+// Python bindings for the dmlcloud_amd gfx950 kernels.
+//
+// This TU is compiled by the host compiler; all kernel launchers live in
+// the .hip TUs and are declared here.
+
+#include <torch/extension.h>
+
+namespace dmlamd {
+
+// reduce.hip
+void metric_reduce_into(at::Tensor value, at::Tensor acc, at::Tensor count, at::Tensor partials,
+                        int64_t op);
+void metric_accumulate_elementwise(at::Tensor value, at::Tensor acc, at::Tensor count,
+                                   int64_t op);
+at::Tensor metric_finalize_dims(at::Tensor acc, std::vector<int64_t> dims, int64_t op);
+
+// copy.hip
+void chunked_copy(at::Tensor units_blob, int64_t nunits);
+
+// optim.hip
+void fused_adam(at::Tensor param, at::Tensor grad, at::Tensor exp_avg, at::Tensor exp_avg_sq,
+                at::Tensor step_t, double lr, double beta1, double beta2, double eps,
+                double weight_decay, double grad_scale);
+void fused_sgd(at::Tensor param, at::Tensor grad, at::Tensor momentum_buf, double lr,
+               double momentum, double weight_decay, double grad_scale, bool use_momentum);
+void l2_norm_and_scale(at::Tensor flat, at::Tensor partials, at::Tensor out, double max_norm,
+                       bool apply);
+
+} // namespace dmlamd
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "dmlcloud_amd gfx950 (MI355X/CDNA4) kernels";
+  m.def("metric_reduce_into", &dmlamd::metric_reduce_into,
+        "Fully reduce value into scalar accumulator (deterministic)");
+  m.def("metric_accumulate_elementwise", &dmlamd::metric_accumulate_elementwise,
+        "Elementwise merge value into accumulator");
+  m.def("metric_finalize_dims", &dmlamd::metric_finalize_dims,
+        "Reduce accumulator over given dims");
+  m.def("chunked_copy", &dmlamd::chunked_copy, "Descriptor-table gather/scatter copy");
+  m.def("fused_adam", &dmlamd::fused_adam, "Fused Adam on flat fp32 buffers");
+  m.def("fused_sgd", &dmlamd::fused_sgd, "Fused SGD on flat fp32 buffers");
+  m.def("l2_norm_and_scale", &dmlamd::l2_norm_and_scale,
+        "Deterministic L2 norm + optional clip scale");
+}

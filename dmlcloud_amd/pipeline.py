@@ -1,0 +1,430 @@
+"""TrainingPipeline: experiment lifecycle orchestration.
+
+API parity with the reference pipeline (reference dmlcloud/pipeline.py:
+20-331): registries for models/optimizers/schedulers/datasets/stages,
+device selection, checkpoint dir setup, wandb, diagnostics, monitored
+barriers over a gloo side group, run loop with cleanup guard.
+
+MI355X-native differences:
+- register_model wraps with RCCL-tuned DDP (parallel/ddp.py: 64 MB
+  buckets sized for 7-way xGMI fan-out, gradient_as_bucket_view) or with
+  the flat-replica fast path (ddp_impl='flat': one flat all-reduce +
+  fused optimizers + hipGraph-capturable steps).
+- the save_latest/save_interval/save_best policy on register_model is
+  actually implemented (checkpoint.ModelCheckpointer + the .dmlt fused
+  pack format), unlike the reference which accepts and ignores it
+  (reference pipeline.py:61-64).
+- device selection maps local_rank -> HIP device; the warning path
+  references HIP_VISIBLE_DEVICES.
+- save_checkpoint()/load_checkpoint() provide full pipeline state
+  round-trip (models, optimizers, schedulers, tracker, stage epochs) on
+  top of the reference's user-hook-only resume.
+"""
+
+import logging
+import warnings
+from datetime import datetime
+from typing import Any, Dict, List, Optional, Sequence, Union
+
+import torch
+import torch.distributed as dist
+from torch.utils.data import DataLoader, Dataset
+
+from .checkpoint import (
+    CheckpointDir,
+    ModelCheckpointer,
+    find_slurm_checkpoint,
+    generate_checkpoint_path,
+    load_tensor_state,
+    save_tensor_state,
+)
+from .config import Config
+from .metrics import MetricTracker, Reduction
+from .parallel.ddp import wrap_ddp
+from .parallel.distributed import all_gather_object, broadcast_object, is_root, local_rank, root_only
+from .parallel.flat import FlatReplica
+from .stage import Stage
+from .utils.logging import IORedirector, add_log_handlers, experiment_header, general_diagnostics
+from .utils.wandb import wandb, wandb_is_initialized, wandb_set_startup_timeout
+
+__all__ = ['TrainingPipeline']
+
+
+class TrainingPipeline:
+    def __init__(self, config: Optional[Union[Config, Dict]] = None, name: Optional[str] = None):
+        self.config = Config.create(config)
+        self.name = name
+
+        self.logger = logging.getLogger('dmlcloud_amd')
+        self.checkpoint_dir = None
+        self.gloo_group = None
+        self.io_redirector = None
+        self.resumed = None
+        self.tracker = MetricTracker()
+        self.device = None
+        self.start_time = None
+        self.stop_time = None
+        self.current_stage = None
+
+        self.wandb = False
+        self._wandb_initializer = None
+
+        self.stages: List[Stage] = []
+        self.datasets = {}
+        self.models = {}
+        self.optimizers = {}
+        self.schedulers = {}
+        self.checkpointers = {}
+
+    @property
+    def checkpointing_enabled(self) -> bool:
+        return self.checkpoint_dir is not None
+
+    # ------------------------------------------------------------ registries
+
+    def register_model(
+        self,
+        name: str,
+        model: torch.nn.Module,
+        use_ddp: bool = True,
+        sync_bn: bool = False,
+        ddp_impl: str = 'torch',
+        bucket_cap_mb: Optional[int] = None,
+        save_latest: bool = True,
+        save_interval: Optional[int] = None,
+        save_best: bool = False,
+        best_metric: str = 'val/loss',
+        verbose: bool = True,
+    ):
+        """Register (and distribute) a model.
+
+        ddp_impl: 'torch' wraps in RCCL-tuned DistributedDataParallel;
+        'flat' uses the flat-replica fast path (fp32 params, explicit
+        grad_sync, fused optimizers, hipGraph-friendly); 'none'/use_ddp=False
+        leaves the module unwrapped (parameters broadcast from rank 0).
+        """
+        if name in self.models:
+            raise ValueError(f'Model with name {name} already exists')
+
+        if use_ddp and ddp_impl == 'torch':
+            model = wrap_ddp(model, self.device, sync_bn=sync_bn, bucket_cap_mb=bucket_cap_mb)
+        elif use_ddp and ddp_impl == 'flat':
+            model = model.to(self.device)
+            if sync_bn:
+                model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
+            model = FlatReplica(model)
+        else:
+            model = model.to(self.device)
+            if dist.is_initialized() and dist.get_world_size() > 1:
+                for t in model.state_dict().values():
+                    if isinstance(t, torch.Tensor):
+                        dist.broadcast(t, src=0)
+        self.models[name] = model
+
+        if self.checkpointing_enabled:
+            self.checkpointers[name] = ModelCheckpointer(
+                self.checkpoint_dir,
+                name,
+                save_latest=save_latest,
+                save_interval=save_interval,
+                save_best=save_best,
+                best_metric=best_metric,
+            )
+
+        if verbose:
+            module = model.module if hasattr(model, 'module') else model
+            n_params = sum(p.numel() for p in module.parameters())
+            msg = f'Model "{name}":\n'
+            msg += f'    - Parameters: {n_params / 1e6:.1f} M\n'
+            msg += f'    - DDP: {use_ddp} ({ddp_impl})\n'
+            msg += f'    - {model}'
+            self.logger.info(msg)
+
+    def register_optimizer(self, name: str, optimizer, scheduler=None):
+        if name in self.optimizers:
+            raise ValueError(f'Optimizer with name {name} already exists')
+        self.optimizers[name] = optimizer
+        if scheduler is not None:
+            self.schedulers[name] = scheduler
+
+    def register_dataset(self, name: str, dataset: Union[DataLoader, Dataset, Sequence], verbose: bool = True):
+        if name in self.datasets:
+            raise ValueError(f'Dataset with name {name} already exists')
+        self.datasets[name] = dataset
+        if verbose:
+            msg = f'Dataset "{name}":\n'
+            try:
+                length = len(dataset)
+                msg += f'    - Batches (Total): ~{length * dist.get_world_size()}\n'
+                msg += f'    - Batches (/Worker): {length}\n'
+            except TypeError:
+                msg += '    - Batches (Total): N/A\n'
+                msg += '    - Batches (/Worker): N/A\n'
+            self.logger.info(msg)
+
+    def append_stage(self, stage: Stage, max_epochs: Optional[int] = None, name: Optional[str] = None):
+        if not isinstance(stage, Stage):
+            raise ValueError('stage must be a Stage object')
+        stage.pipeline = self
+        stage.max_epochs = max_epochs
+        stage.name = name
+        self.stages.append(stage)
+
+    # --------------------------------------------------------- checkpointing
+
+    def enable_checkpointing(self, root: str, resume: bool = False):
+        if self.checkpointing_enabled:
+            raise ValueError('Checkpointing already enabled')
+
+        path = None
+        if resume and CheckpointDir(root).is_valid:
+            path = root
+            self.resumed = True
+        elif resume and find_slurm_checkpoint(root):
+            path = find_slurm_checkpoint(root)
+            self.resumed = True
+
+        if path is None:  # dir creation happens later, in _pre_run
+            path = generate_checkpoint_path(root=root, name=self.name, creation_time=self.start_time)
+            if dist.is_initialized():
+                path = broadcast_object(path)
+            self.resumed = False
+
+        self.checkpoint_dir = CheckpointDir(path)
+
+    def save_checkpoint(self):
+        """Save full pipeline state (models, optimizers, schedulers, metric
+        tracker, stage epochs) to the checkpoint dir. Root only."""
+        if not self.checkpointing_enabled:
+            raise ValueError('Checkpointing is not enabled')
+        if not is_root():
+            return
+        state: Dict[str, Any] = {
+            'models': {},
+            'optimizers': {},
+            'schedulers': {},
+            'tracker': self.tracker.state_dict(),
+            'stages': [stage.current_epoch for stage in self.stages],
+        }
+        for name, model in self.models.items():
+            module = model.module if hasattr(model, 'module') else model
+            state['models'][name] = module.state_dict()
+        for name, opt in self.optimizers.items():
+            state['optimizers'][name] = opt.state_dict()
+        for name, sched in self.schedulers.items():
+            state['schedulers'][name] = sched.state_dict()
+        save_tensor_state(state, self.checkpoint_dir.state_file)
+
+    def load_checkpoint(self):
+        """Restore pipeline state saved by save_checkpoint (all ranks)."""
+        if not self.checkpointing_enabled:
+            raise ValueError('Checkpointing is not enabled')
+        state = load_tensor_state(self.checkpoint_dir.state_file, device=self.device)
+        for name, sd in state['models'].items():
+            if name in self.models:
+                model = self.models[name]
+                module = model.module if hasattr(model, 'module') else model
+                module.load_state_dict(sd)
+        for name, sd in state['optimizers'].items():
+            if name in self.optimizers:
+                self.optimizers[name].load_state_dict(sd)
+        for name, sd in state['schedulers'].items():
+            if name in self.schedulers:
+                self.schedulers[name].load_state_dict(sd)
+        self.tracker.load_state_dict(state['tracker'])
+        for stage, epoch in zip(self.stages, state.get('stages', [])):
+            stage.current_epoch = epoch
+        return state
+
+    # ----------------------------------------------------------------- wandb
+
+    def enable_wandb(
+        self,
+        project: Optional[str] = None,
+        entity: Optional[str] = None,
+        group: Optional[str] = None,
+        tags: Optional[List[str]] = None,
+        startup_timeout: int = 360,
+        **kwargs,
+    ):
+        import wandb as _wandb  # early import to surface availability problems
+
+        @root_only
+        def initializer():
+            wandb_set_startup_timeout(startup_timeout)
+            _wandb.init(
+                config=self.config.to_container(resolve=True),
+                name=self.name,
+                entity=entity,
+                project=project if project else self.name,
+                group=group,
+                tags=tags,
+                **kwargs,
+            )
+
+        self._wandb_initializer = initializer
+        self.wandb = True
+
+    # -------------------------------------------------------------- tracking
+
+    def track_reduce(
+        self,
+        name: str,
+        value: torch.Tensor,
+        step: Optional[int] = None,
+        reduction: Reduction = Reduction.MEAN,
+        dim: Optional[List[int]] = None,
+        reduce_globally: bool = True,
+    ):
+        if name not in self.tracker:
+            self.tracker.register_metric(name, reduction, dim, reduce_globally)
+        self.tracker.track(name, value)
+
+    def track(self, name: str, value: Any, step: Optional[int] = None):
+        if name not in self.tracker:
+            self.tracker.register_metric(name)
+        self.tracker.track(name, value)
+
+    # ------------------------------------------------------------------- run
+
+    def barrier(self, timeout=None):
+        from datetime import timedelta
+
+        if self.gloo_group is None:
+            dist.barrier()
+        else:
+            td = timedelta(seconds=timeout) if timeout is not None else None
+            dist.monitored_barrier(self.gloo_group, timeout=td, wait_all_ranks=True)
+
+    def run(self):
+        """Run all registered stages."""
+        with _RunGuard(self):
+            self._pre_run()
+            for stage in self.stages:
+                self.current_stage = stage
+                stage.run()
+            self._post_run()
+
+    def pre_run(self):
+        pass
+
+    def post_run(self):
+        pass
+
+    def resume_run(self):
+        pass
+
+    def _pre_run(self):
+        if len(self.stages) == 0:
+            raise ValueError('No stages defined. Use append_stage() to add stages to the pipeline.')
+        if not dist.is_initialized():
+            raise ValueError(
+                'Default process group not initialized! Call init_process_group_auto() or '
+                'torch.distributed.init_process_group() first.'
+            )
+
+        if dist.is_gloo_available():
+            self.gloo_group = dist.new_group(backend='gloo')
+        else:
+            warnings.warn('Gloo backend not available. Barriers will not use custom timeouts.')
+
+        if torch.cuda.is_available():
+            if local_rank() is None:
+                warnings.warn(
+                    'GPU is available but no local rank found. Make sure to set HIP_VISIBLE_DEVICES '
+                    'manually for each rank.'
+                )
+                self.device = torch.device('cuda')
+            else:
+                self.device = torch.device('cuda', local_rank())
+                torch.cuda.set_device(local_rank())
+        else:
+            warnings.warn('No GPU available. Running on CPU.')
+            self.device = torch.device('cpu')
+
+        # prevent checkpoint dir creation before all ranks searched for it
+        self.barrier(timeout=10 * 60)
+        if self.checkpointing_enabled:
+            self._init_checkpointing()
+
+        if self.wandb:
+            self._wandb_initializer()
+
+        self.barrier(timeout=10 * 60)
+        self.start_time = datetime.now()
+
+        add_log_handlers(self.logger)
+        header = '\n' + experiment_header(self.name, self.checkpoint_dir, self.start_time)
+        self.logger.info(header)
+
+        if self.resumed:
+            self._resume_run()
+
+        diagnostics = general_diagnostics()
+        diagnostics += '\n* DEVICES:\n'
+        devices = all_gather_object(str(self.device))
+        diagnostics += '\n'.join(f'    - [Rank {i}] {device}' for i, device in enumerate(devices))
+        diagnostics += '\n* CONFIG:\n'
+        diagnostics += '\n'.join(f'    {line}' for line in self.config.to_yaml(resolve=True).splitlines())
+        self.logger.info(diagnostics)
+
+        self.pre_run()
+
+    @root_only
+    def _init_checkpointing(self):
+        if not self.checkpoint_dir.is_valid:
+            self.checkpoint_dir.create()
+            self.checkpoint_dir.save_config(self.config)
+        self.io_redirector = IORedirector(self.checkpoint_dir.log_file)
+        self.io_redirector.install()
+
+    def _resume_run(self):
+        self.logger.info(f'Resuming training from checkpoint: {self.checkpoint_dir}')
+        self.resume_run()
+
+    def _post_run(self):
+        self.stop_time = datetime.now()
+        self.logger.info(f'Finished training in {self.stop_time - self.start_time} ({self.stop_time})')
+        if self.checkpointing_enabled:
+            self.logger.info(f'Outputs have been saved to {self.checkpoint_dir}')
+        self.post_run()
+
+    def _pre_epoch(self):
+        pass
+
+    def _post_epoch(self):
+        if self.wandb and is_root():
+            metrics = {name: self.tracker[name][-1] for name in self.tracker if self.tracker[name]}
+            wandb.log(metrics)
+        if self.checkpointing_enabled and is_root() and self.current_stage is not None:
+            epoch = self.current_stage.current_epoch
+            for name, checkpointer in self.checkpointers.items():
+                checkpointer.maybe_save(self.models[name], epoch, tracker=self.tracker)
+
+    def _cleanup(self, exc_type, exc_value, traceback):
+        """Called by _RunGuard so failures still flush wandb and stdio."""
+        if exc_type is KeyboardInterrupt:
+            self.logger.info('------- Training interrupted by user -------')
+        elif exc_type is not None:
+            self.logger.error(
+                '------- Training failed with an exception -------', exc_info=(exc_type, exc_value, traceback)
+            )
+
+        if self.wandb and wandb_is_initialized():
+            wandb.finish(exit_code=0 if exc_type is None else 1)
+
+        if self.io_redirector is not None:
+            self.io_redirector.uninstall()
+
+        return False
+
+
+class _RunGuard:
+    def __init__(self, pipeline):
+        self.pipeline = pipeline
+
+    def __enter__(self):
+        pass
+
+    def __exit__(self, exc_type, exc_value, traceback):
+        return self.pipeline._cleanup(exc_type, exc_value, traceback)

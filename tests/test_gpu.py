@@ -1,0 +1,307 @@
+"""gfx950 kernel numerics and GPU end-to-end paths.
+
+Every test compares the HIP kernels against the pure-torch reference
+implementations (ops/_reference.py) or against plain fp32 torch ops.
+Run on an MI355X box: python -m pytest tests -m gpu
+"""
+
+import sys
+
+import pytest
+import torch
+
+from dmlcloud_amd import ops
+from dmlcloud_amd.ops import OP_MAX, OP_MIN, OP_SUM
+from dmlcloud_amd.ops import _reference as ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = 'cuda:0'
+
+
+def _assert_native_loaded():
+    assert ops.is_available(), 'native extension must be present on a GPU box'
+
+
+class TestReduceKernels:
+    @pytest.mark.parametrize('op', [OP_SUM, OP_MIN, OP_MAX])
+    @pytest.mark.parametrize('n', [1, 7, 256, 16384, 16385, 1 << 20])
+    @pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16, torch.float16])
+    def test_reduce_into_float(self, op, n, dtype):
+        _assert_native_loaded()
+        torch.manual_seed(n)
+        value = torch.randn(n, dtype=torch.float32)
+        value_dev = value.to(DEV, dtype)
+
+        acc_ref = torch.zeros(1, dtype=torch.float64)
+        cnt_ref = torch.zeros(1, dtype=torch.int64)
+        ref.reduce_into_acc(value.to(dtype), acc_ref, cnt_ref, op)
+
+        acc = torch.zeros(1, dtype=torch.float64, device=DEV)
+        if op == OP_MIN:
+            acc.fill_(float('inf'))
+            acc_ref2 = torch.full((1,), float('inf'), dtype=torch.float64)
+            ref.reduce_into_acc(value.to(dtype), acc_ref2, torch.zeros(1, dtype=torch.int64), op)
+            acc_ref = acc_ref2
+        elif op == OP_MAX:
+            acc.fill_(float('-inf'))
+            acc_ref2 = torch.full((1,), float('-inf'), dtype=torch.float64)
+            ref.reduce_into_acc(value.to(dtype), acc_ref2, torch.zeros(1, dtype=torch.int64), op)
+            acc_ref = acc_ref2
+        cnt = torch.zeros(1, dtype=torch.int64, device=DEV)
+        ops.metric_reduce_into(value_dev, acc, cnt, op)
+
+        assert cnt.item() == 1
+        torch.testing.assert_close(acc.cpu(), acc_ref, rtol=1e-10, atol=1e-8)
+
+    @pytest.mark.parametrize('op', [OP_SUM, OP_MIN, OP_MAX])
+    def test_reduce_into_int64(self, op):
+        _assert_native_loaded()
+        value = torch.randint(-1000, 1000, (4096,), dtype=torch.int64)
+        init = {OP_SUM: 0, OP_MIN: torch.iinfo(torch.int64).max, OP_MAX: torch.iinfo(torch.int64).min}[op]
+        acc = torch.full((1,), init, dtype=torch.int64, device=DEV)
+        cnt = torch.zeros(1, dtype=torch.int64, device=DEV)
+        ops.metric_reduce_into(value.to(DEV), acc, cnt, op)
+        expected = {OP_SUM: value.sum(), OP_MIN: value.min(), OP_MAX: value.max()}[op]
+        assert acc.item() == expected.item()
+
+    def test_deterministic(self):
+        """Same input -> bitwise-identical accumulator (no atomics)."""
+        _assert_native_loaded()
+        value = torch.randn(1 << 20, device=DEV)
+        results = []
+        for _ in range(3):
+            acc = torch.zeros(1, dtype=torch.float64, device=DEV)
+            cnt = torch.zeros(1, dtype=torch.int64, device=DEV)
+            ops.metric_reduce_into(value, acc, cnt, OP_SUM)
+            results.append(acc.item())
+        assert results[0] == results[1] == results[2]
+
+    @pytest.mark.parametrize('op', [OP_SUM, OP_MIN, OP_MAX])
+    def test_elementwise_accumulate(self, op):
+        _assert_native_loaded()
+        torch.manual_seed(0)
+        values = [torch.randn(4, 5) for _ in range(3)]
+
+        acc_ref = torch.zeros(4, 5, dtype=torch.float64)
+        cnt_ref = torch.zeros(1, dtype=torch.int64)
+        acc_dev = torch.zeros(4, 5, dtype=torch.float64, device=DEV)
+        cnt_dev = torch.zeros(1, dtype=torch.int64, device=DEV)
+        if op != OP_SUM:
+            fill = float('inf') if op == OP_MIN else float('-inf')
+            acc_ref.fill_(fill)
+            acc_dev.fill_(fill)
+        for v in values:
+            ref.accumulate_elementwise(v, acc_ref, cnt_ref, op)
+            ops.metric_accumulate_elementwise(v.to(DEV), acc_dev, cnt_dev, op)
+        assert cnt_dev.item() == 3
+        torch.testing.assert_close(acc_dev.cpu(), acc_ref)
+
+    @pytest.mark.parametrize('op', [OP_SUM, OP_MIN, OP_MAX])
+    @pytest.mark.parametrize('dims', [[0], [1], [0, 2], [1, 2], [0, 1, 2]])
+    def test_finalize_dims(self, op, dims):
+        _assert_native_loaded()
+        torch.manual_seed(1)
+        acc = torch.randn(3, 4, 5, dtype=torch.float64)
+        out_ref = ref.finalize_dims(acc, dims, op)
+        out_dev = ops.metric_finalize_dims(acc.to(DEV), dims, op)
+        torch.testing.assert_close(out_dev.cpu(), out_ref)
+
+
+class TestMetricsEndToEndGPU:
+    def test_reducer_parity_with_cpu(self):
+        """Same appended values -> same result on GPU accumulators as on
+        the CPU torch path."""
+        from dmlcloud_amd.metrics import MetricReducer, Reduction
+
+        torch.manual_seed(0)
+        values = [torch.randn(128) for _ in range(10)]
+        for reduction in [Reduction.MEAN, Reduction.SUM, Reduction.MIN, Reduction.MAX]:
+            r_cpu = MetricReducer(reduction)
+            r_gpu = MetricReducer(reduction)
+            for v in values:
+                r_cpu.append(v)
+                r_gpu.append(v.to(DEV))
+            a = r_cpu.reduce_locally()
+            b = r_gpu.reduce_locally().cpu()
+            torch.testing.assert_close(a, b, rtol=1e-9, atol=1e-9)
+
+    def test_tracker_epoch_gpu(self, torch_distributed_cuda):
+        from dmlcloud_amd.metrics import MetricTracker, Reduction
+
+        t = MetricTracker()
+        t.register_metric('loss', Reduction.MEAN)
+        for i in range(5):
+            t.track('loss', torch.full((1,), float(i), device=DEV))
+        t.next_epoch()
+        assert t['loss'][0].item() == pytest.approx(2.0)
+
+
+class TestOptimKernels:
+    def test_adam_matches_reference(self):
+        _assert_native_loaded()
+        torch.manual_seed(0)
+        n = (1 << 16) + 3  # odd tail exercises the scalar path
+        param_ref = torch.randn(n)
+        param_dev = param_ref.to(DEV)
+        ea_ref, eas_ref = torch.zeros(n), torch.zeros(n)
+        ea_dev, eas_dev = torch.zeros(n, device=DEV), torch.zeros(n, device=DEV)
+        st_ref = torch.zeros(1, dtype=torch.int32)
+        st_dev = torch.zeros(1, dtype=torch.int32, device=DEV)
+        for it in range(3):
+            torch.manual_seed(it)
+            grad = torch.randn(n)
+            ref.fused_adam_step(param_ref, grad, ea_ref, eas_ref, st_ref, 1e-2, 0.9, 0.999, 1e-8, 0.01, 0.5)
+            ops.fused_adam(param_dev, grad.to(DEV), ea_dev, eas_dev, st_dev, 1e-2, 0.9, 0.999, 1e-8, 0.01, 0.5)
+        torch.testing.assert_close(param_dev.cpu(), param_ref, rtol=1e-4, atol=1e-6)
+
+    def test_sgd_matches_reference(self):
+        _assert_native_loaded()
+        torch.manual_seed(0)
+        n = 12345
+        param_ref = torch.randn(n)
+        param_dev = param_ref.to(DEV)
+        mom_ref = torch.zeros(n)
+        mom_dev = torch.zeros(n, device=DEV)
+        for it in range(3):
+            torch.manual_seed(10 + it)
+            grad = torch.randn(n)
+            ref.fused_sgd_step(param_ref, grad, mom_ref, 0.1, 0.9, 0.001, 1.0)
+            ops.fused_sgd(param_dev, grad.to(DEV), mom_dev, 0.1, 0.9, 0.001, 1.0)
+        torch.testing.assert_close(param_dev.cpu(), param_ref, rtol=1e-5, atol=1e-6)
+
+    def test_clip_matches_torch(self):
+        _assert_native_loaded()
+        torch.manual_seed(0)
+        g = torch.randn(1 << 20, device=DEV)
+        g_ref = g.cpu().clone()
+        norm = ops.clip_grad_norm_(g, 0.5)
+        p = torch.nn.Parameter(torch.zeros_like(g_ref))
+        p.grad = g_ref
+        norm_ref = torch.nn.utils.clip_grad_norm_([p], 0.5)
+        assert norm.item() == pytest.approx(norm_ref.item(), rel=1e-5)
+        torch.testing.assert_close(g.cpu(), p.grad, rtol=1e-5, atol=1e-7)
+
+
+class TestCopyKernel:
+    def test_pack_unpack_roundtrip(self):
+        _assert_native_loaded()
+        torch.manual_seed(0)
+        tensors = [torch.randn(n, device=DEV) for n in (1024, 17, 1 << 20)]
+        total = sum(t.numel() for t in tensors)
+        flat = torch.zeros(total, device=DEV)
+        offs = []
+        off = 0
+        for t in tensors:
+            offs.append(off)
+            off += t.numel()
+        ops.chunked_copy(tensors, [flat[o : o + t.numel()] for o, t in zip(offs, tensors)])
+        back = [torch.zeros_like(t) for t in tensors]
+        ops.chunked_copy([flat[o : o + t.numel()] for o, t in zip(offs, tensors)], back)
+        torch.cuda.synchronize()
+        for t, b in zip(tensors, back):
+            torch.testing.assert_close(t, b)
+
+    def test_interleave_gpu_matches_cpu(self):
+        from dmlcloud_amd.data import interleave_batches
+
+        torch.manual_seed(0)
+        batches = [torch.randn(64, 7) for _ in range(4)]
+        cpu_out = [b.clone() for b in interleave_batches(iter(batches), 4)]
+        gpu_out = [b.cpu().clone() for b in interleave_batches(iter([b.to(DEV) for b in batches]), 4)]
+        for a, b in zip(cpu_out, gpu_out):
+            torch.testing.assert_close(a, b)
+
+    def test_dmlt_gpu_roundtrip(self, tmp_path):
+        from dmlcloud_amd.checkpoint import load_tensor_state, save_tensor_state
+
+        state = {'w': torch.randn(1000, 10, device=DEV), 'b': torch.randn(10, device=DEV), 'meta': 3}
+        save_tensor_state(state, tmp_path / 's.dmlt')
+        loaded = load_tensor_state(tmp_path / 's.dmlt', device=DEV)
+        assert loaded['w'].device.type == 'cuda'
+        torch.testing.assert_close(loaded['w'], state['w'])
+        torch.testing.assert_close(loaded['b'], state['b'])
+
+
+class TestFlatReplicaGPU:
+    def test_train_convergence(self, torch_distributed_cuda):
+        from dmlcloud_amd.parallel import FlatAdam, FlatReplica
+
+        torch.manual_seed(0)
+        model = torch.nn.Sequential(torch.nn.Linear(16, 64), torch.nn.ReLU(), torch.nn.Linear(64, 1)).to(DEV)
+        replica = FlatReplica(model)
+        opt = FlatAdam(replica, lr=1e-2)
+        x = torch.randn(256, 16, device=DEV)
+        y = x.sum(dim=1, keepdim=True)
+        first_loss = None
+        for _ in range(50):
+            replica.zero_grad()
+            loss = torch.nn.functional.mse_loss(replica(x), y)
+            loss.backward()
+            replica.grad_sync()
+            opt.step()
+            if first_loss is None:
+                first_loss = loss.item()
+        assert loss.item() < first_loss * 0.1
+
+    def test_graphed_step(self, torch_distributed_cuda):
+        from dmlcloud_amd.parallel import FlatReplica, FlatSGD, GraphedStep
+
+        torch.manual_seed(0)
+        model = torch.nn.Linear(32, 32).to(DEV)
+        replica = FlatReplica(model)
+        opt = FlatSGD(replica, lr=1e-3)
+        x = torch.randn(64, 32, device=DEV)
+
+        def step():
+            replica.zero_grad()
+            loss = replica(x).pow(2).mean()
+            loss.backward()
+            replica.grad_sync()
+            opt.step()
+
+        gs = GraphedStep(step, warmup=3)
+        gs.initialize()
+        assert gs.captured, 'hipGraph capture must succeed for the flat path'
+        p0 = replica.flat_param.clone()
+        for _ in range(5):
+            gs()
+        torch.cuda.synchronize()
+        assert not torch.equal(p0, replica.flat_param)
+
+
+class TestPipelineGPU:
+    def test_smoke_gpu(self, torch_distributed_cuda):
+        from dmlcloud_amd import TrainingPipeline, TrainValStage
+
+        class DS(torch.utils.data.Dataset):
+            def __len__(self):
+                return 16
+
+            def __getitem__(self, idx):
+                g = torch.Generator().manual_seed(idx)
+                return torch.randn(10, generator=g), idx % 10
+
+        class Stage_(TrainValStage):
+            def pre_stage(self):
+                model = torch.nn.Linear(10, 10)
+                self.pipeline.register_model('m', model)
+                self.pipeline.register_optimizer('sgd', torch.optim.SGD(model.parameters(), lr=1e-2))
+                self.pipeline.register_dataset('train', torch.utils.data.DataLoader(DS(), batch_size=4))
+                self.pipeline.register_dataset('val', torch.utils.data.DataLoader(DS(), batch_size=4))
+                self.loss = torch.nn.CrossEntropyLoss()
+
+            def step(self, batch):
+                x, y = batch
+                x, y = x.to(self.device), y.to(self.device)
+                return self.loss(self.pipeline.models['m'](x), y)
+
+        pipeline = TrainingPipeline()
+        pipeline.append_stage(Stage_(), max_epochs=2)
+        pipeline.run()
+        assert pipeline.device.type == 'cuda'
+        assert pipeline.tracker['train/loss'][1] is not None
+
+
+if __name__ == '__main__':
+    sys.exit(pytest.main([__file__, '-m', 'gpu']))
