@@ -53,6 +53,8 @@ def parse_args():
     p.add_argument('--impl', choices=['flat', 'ddp'], default='flat')
     p.add_argument('--batch-size', type=int, default=None, help='per-GPU batch size')
     p.add_argument('--no-graph', action='store_true', help='disable hipGraph capture')
+    p.add_argument('--no-fused', action='store_true', help='disable the fused smallcnn kernels (MIOpen path)')
+    p.add_argument('--channels-last', action='store_true', help='NHWC memory format (resnet50)')
     p.add_argument('--seq-len', type=int, default=1024, help='gpt2 sequence length')
     return p.parse_args()
 
@@ -76,11 +78,19 @@ class BenchStage(TrainValStage):
         torch.manual_seed(1234)
 
         if args.model == 'mnist':
-            model = mnist_cnn()
+            if device.type == 'cuda' and not args.no_fused:
+                from dmlcloud_amd.ops.fused_cnn import FusedMnistCNN
+
+                model = FusedMnistCNN()  # fused gfx950 conv+relu+pool kernels
+            else:
+                model = mnist_cnn()
             self.batch_shape = (args.batch_size, 1, 28, 28)
             self.dtype = 'fp32'
         elif args.model == 'resnet50':
             model = resnet50()
+            if args.channels_last:
+                model = model.to(memory_format=torch.channels_last)
+                torch.backends.cudnn.benchmark = True
             self.batch_shape = (args.batch_size, 3, 224, 224)
             self.dtype = 'bf16'
         else:  # gpt2
@@ -91,10 +101,10 @@ class BenchStage(TrainValStage):
         if args.impl == 'flat':
             self.pipeline.register_model('net', model, ddp_impl='flat', verbose=False)
             replica = self.pipeline.models['net']
-            if args.model == 'mnist':
-                self.pipeline.register_optimizer('opt', FlatAdam(replica, lr=1e-3))
-            else:
+            if args.model == 'resnet50':
                 self.pipeline.register_optimizer('opt', FlatSGD(replica, lr=1e-3, momentum=0.9))
+            else:
+                self.pipeline.register_optimizer('opt', FlatAdam(replica, lr=1e-3))
         else:
             use_ddp = dist.is_initialized() and dist.get_world_size() > 1
             self.pipeline.register_model('net', model, use_ddp=use_ddp, verbose=False)
@@ -114,7 +124,11 @@ class BenchStage(TrainValStage):
             self.labels = None
             self.static_batch = torch.zeros_like(self.pool[0])
         else:
-            self.pool = [torch.randn(self.batch_shape, generator=g).to(device) for _ in range(self.n_pool)]
+            mf = torch.channels_last if (args.model == 'resnet50' and args.channels_last) else torch.contiguous_format
+            self.pool = [
+                torch.randn(self.batch_shape, generator=g).to(device).to(memory_format=mf)
+                for _ in range(self.n_pool)
+            ]
             self.labels = [
                 torch.randint(0, 10 if args.model == 'mnist' else 1000, (args.batch_size,), generator=g).to(device)
                 for _ in range(self.n_pool)

@@ -26,6 +26,14 @@ void fused_sgd(at::Tensor param, at::Tensor grad, at::Tensor momentum_buf, doubl
 void l2_norm_and_scale(at::Tensor flat, at::Tensor partials, at::Tensor out, double max_norm,
                        bool apply);
 
+// smallcnn.hip
+void conv3x3_relu_pool_fwd(at::Tensor in, at::Tensor w, at::Tensor b, at::Tensor out,
+                           at::Tensor argmax);
+void conv3x3_relu_pool_bwd_data(at::Tensor dpooled, at::Tensor argmax, at::Tensor pooled,
+                                at::Tensor w, at::Tensor din);
+void conv3x3_relu_pool_bwd_weight(at::Tensor dpooled, at::Tensor argmax, at::Tensor pooled,
+                                  at::Tensor in, at::Tensor dw, at::Tensor db);
+
 } // namespace dmlamd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -41,4 +49,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &dmlamd::fused_sgd, "Fused SGD on flat fp32 buffers");
   m.def("l2_norm_and_scale", &dmlamd::l2_norm_and_scale,
         "Deterministic L2 norm + optional clip scale");
+  m.def("conv3x3_relu_pool_fwd", &dmlamd::conv3x3_relu_pool_fwd,
+        "Fused conv3x3(pad1)+ReLU+maxpool2x2 forward");
+  m.def("conv3x3_relu_pool_bwd_data", &dmlamd::conv3x3_relu_pool_bwd_data,
+        "Fused conv+relu+pool input gradient");
+  m.def("conv3x3_relu_pool_bwd_weight", &dmlamd::conv3x3_relu_pool_bwd_weight,
+        "Fused conv+relu+pool weight/bias gradient");
 }

@@ -23,6 +23,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, 'reduce.hip'),
         os.path.join(CSRC, 'copy.hip'),
         os.path.join(CSRC, 'optim.hip'),
+        os.path.join(CSRC, 'smallcnn.hip'),
     ],
     extra_compile_args={
         'cxx': ['-O3', '-std=c++17'],

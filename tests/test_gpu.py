@@ -179,8 +179,10 @@ class TestOptimKernels:
         p = torch.nn.Parameter(torch.zeros_like(g_ref))
         p.grad = g_ref
         norm_ref = torch.nn.utils.clip_grad_norm_([p], 0.5)
-        assert norm.item() == pytest.approx(norm_ref.item(), rel=1e-5)
-        torch.testing.assert_close(g.cpu(), p.grad, rtol=1e-5, atol=1e-7)
+        # the gfx950 kernel accumulates in fp64; torch's fp32 norm carries
+        # ~1e-5 relative accumulation error at 1M elements
+        assert norm.item() == pytest.approx(norm_ref.item(), rel=1e-4)
+        torch.testing.assert_close(g.cpu(), p.grad, rtol=1e-4, atol=1e-7)
 
 
 class TestCopyKernel:
@@ -268,6 +270,73 @@ class TestFlatReplicaGPU:
             gs()
         torch.cuda.synchronize()
         assert not torch.equal(p0, replica.flat_param)
+
+
+class TestFusedCnn:
+    """Fused conv3x3+relu+pool kernels vs the plain torch fp32 oracle."""
+
+    @pytest.mark.parametrize('cin,cout,hw,n', [(1, 16, 28, 8), (16, 16, 14, 8), (3, 32, 32, 4), (5, 7, 12, 3)])
+    def test_forward_matches_torch(self, cin, cout, hw, n):
+        import torch.nn.functional as F
+
+        from dmlcloud_amd.ops.fused_cnn import ConvReluPool2d
+
+        torch.manual_seed(0)
+        layer = ConvReluPool2d(cin, cout).to(DEV)
+        x = torch.randn(n, cin, hw, hw, device=DEV)
+        out = layer(x)
+        ref_out = F.max_pool2d(F.relu(F.conv2d(x, layer.weight, layer.bias, padding=1)), 2)
+        torch.testing.assert_close(out, ref_out, rtol=1e-5, atol=1e-5)
+
+    @pytest.mark.parametrize('cin,cout,hw,n', [(1, 16, 28, 4), (16, 16, 14, 4), (4, 8, 8, 2)])
+    def test_backward_matches_torch(self, cin, cout, hw, n):
+        import torch.nn.functional as F
+
+        from dmlcloud_amd.ops.fused_cnn import ConvReluPool2d
+
+        torch.manual_seed(1)
+        layer = ConvReluPool2d(cin, cout).to(DEV)
+        x = torch.randn(n, cin, hw, hw, device=DEV, requires_grad=True)
+        out = layer(x)
+        g = torch.randn_like(out)
+        out.backward(g)
+
+        w2 = layer.weight.detach().clone().requires_grad_(True)
+        b2 = layer.bias.detach().clone().requires_grad_(True)
+        x2 = x.detach().clone().requires_grad_(True)
+        ref = F.max_pool2d(F.relu(F.conv2d(x2, w2, b2, padding=1)), 2)
+        ref.backward(g)
+
+        torch.testing.assert_close(layer.weight.grad, w2.grad, rtol=1e-4, atol=1e-4)
+        torch.testing.assert_close(layer.bias.grad, b2.grad, rtol=1e-4, atol=1e-4)
+        torch.testing.assert_close(x.grad, x2.grad, rtol=1e-4, atol=1e-4)
+
+    def test_full_model_matches_eager(self):
+        """FusedMnistCNN step == plain-torch mnist_cnn step (same weights)."""
+        from dmlcloud_amd.models import mnist_cnn
+        from dmlcloud_amd.ops.fused_cnn import FusedMnistCNN
+
+        torch.manual_seed(0)
+        fused = FusedMnistCNN().to(DEV)
+        eager = mnist_cnn().to(DEV)
+        with torch.no_grad():
+            eager[0].weight.copy_(fused.layer1.weight)
+            eager[0].bias.copy_(fused.layer1.bias)
+            eager[3].weight.copy_(fused.layer2.weight)
+            eager[3].bias.copy_(fused.layer2.bias)
+            eager[7].weight.copy_(fused.fc.weight)
+            eager[7].bias.copy_(fused.fc.bias)
+
+        x = torch.randn(32, 1, 28, 28, device=DEV)
+        y = torch.randint(0, 10, (32,), device=DEV)
+        loss_f = torch.nn.functional.cross_entropy(fused(x), y)
+        loss_e = torch.nn.functional.cross_entropy(eager(x), y)
+        torch.testing.assert_close(loss_f, loss_e, rtol=1e-5, atol=1e-5)
+
+        loss_f.backward()
+        loss_e.backward()
+        torch.testing.assert_close(fused.layer1.weight.grad, eager[0].weight.grad, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(fused.layer2.weight.grad, eager[3].weight.grad, rtol=1e-4, atol=1e-5)
 
 
 class TestPipelineGPU:
