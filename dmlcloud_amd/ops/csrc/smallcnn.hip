@@ -4,27 +4,19 @@
 // the benchmark MNIST-CNN as im2col + hundreds of small batched GEMMs per
 // step plus a pathological find phase. The model's tensors are tiny, so
 // the MI355X-native design stages whole per-sample planes in LDS and
-// computes each layer in ONE kernel per direction:
+// computes each layer in ONE kernel per direction.
 //
-//   forward     — one workgroup per sample: the input plane (CIN x H x W)
-//                 and the weights are loaded into LDS once; every conv+
-//                 relu+pool output of the sample is computed from LDS.
-//                 Global traffic = read input once, write pooled+argmax
-//                 once (the theoretical minimum).
-//   bwd_data    — one workgroup per sample: the sparse conv-gradient
-//                 plane (relu+maxpool gradient folded in via the saved
-//                 argmax) is materialized in LDS, then correlated with
-//                 the flipped weights from LDS.
-//   bwd_weight  — one workgroup per S-sample chunk: samples' input and
-//                 gradient planes staged in LDS; each thread owns one
-//                 (cout,cin) pair slice and accumulates its 9 taps in
-//                 registers; cross-slice LDS tree; one global atomicAdd
-//                 per output per workgroup.
-//
-// A first version of these kernels used one thread per output with global
-// scalar loads: numerically identical but ~64x cache-amplified and
-// latency-bound — slower than MIOpen. The LDS-staged structure below is
-// the fix (guide §2/§6: stage reused tiles in LDS, coalesce global).
+// Iteration history (each measured with rocprofv3 on MI355X):
+//   v1  one thread per output, global scalar loads       -> cache-amplified,
+//       slower than MIOpen.
+//   v2  LDS-staged per-sample workgroups                 -> 3x faster, but
+//       bwd_weight dominated (global atomics per 2-sample block) and the
+//       in-loop bounds checks serialized the tap loops.
+//   v3  (this file) adds: halo-padded LDS planes so every 3x3/4x4 tap loop
+//       is branch-free; +4-float channel strides so simultaneous
+//       cross-channel LDS reads land on different banks; bwd_weight loops
+//       a multi-sample chunk inside one block (register accumulators live
+//       across the chunk), cutting global atomics by the chunk factor.
 //
 // ReLU/maxpool tie-breaking matches torch: first index wins ties, and a
 // pooled value of exactly 0 (all-negative window) propagates no gradient.
@@ -36,7 +28,29 @@
 
 namespace dmlamd {
 
-// Cooperative coalesced copy of `count` floats global->LDS.
+// Padded channel stride: (H+2) halo rows x (W+2) halo cols, plus 4 floats
+// so that simultaneous reads at equal (y,x) across channels hit different
+// LDS banks ((H+2)*(W+2) is typically a multiple of 64).
+__host__ __device__ __forceinline__ int padded_cstride(int H, int W) {
+  return (H + 2) * (W + 2) + 4;
+}
+
+// Zero an LDS range then fill its interior with one input plane (halo stays 0).
+__device__ __forceinline__ void stage_plane_padded(const float* __restrict__ g,
+                                                   float* __restrict__ lds, int CIN, int H,
+                                                   int W) {
+  const int cs = padded_cstride(H, W), PADW = W + 2;
+  for (int i = threadIdx.x; i < CIN * cs; i += kBlock) lds[i] = 0.0f;
+  __syncthreads();
+  const int plane = H * W;
+  for (int i = threadIdx.x; i < CIN * plane; i += kBlock) {
+    const int ci = i / plane;
+    const int rem = i - ci * plane;
+    const int y = rem / W, x = rem - y * W;
+    lds[ci * cs + (y + 1) * PADW + (x + 1)] = g[i];
+  }
+}
+
 __device__ __forceinline__ void stage_to_lds(const float* __restrict__ g, float* __restrict__ l,
                                              int count) {
   for (int i = threadIdx.x; i < count; i += kBlock) l[i] = g[i];
@@ -44,44 +58,43 @@ __device__ __forceinline__ void stage_to_lds(const float* __restrict__ g, float*
 
 // --------------------------------------------------------------- forward
 
-// Workgroup = one sample. LDS: [CIN*H*W] input plane + [COUT*CIN*9] weights.
+// Workgroup = one sample (grid-stride over samples).
+// LDS: [CIN][cstride] halo-padded input plane + [COUT*CIN*9] weights.
 __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_fwd_kernel(
     const float* __restrict__ in, const float* __restrict__ w, const float* __restrict__ bias,
     float* __restrict__ out, uint8_t* __restrict__ argmax, int N, int CIN, int COUT, int H,
     int W) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* ilds = (float*)smem; // [CIN][H][W]
-  float* wlds = ilds + CIN * H * W; // [COUT][CIN][9]
+  const int cs = padded_cstride(H, W), PADW = W + 2;
+  float* ilds = (float*)smem; // [CIN][cs]
+  float* wlds = ilds + CIN * cs; // [COUT][CIN][9]
 
   const int plane = CIN * H * W;
   const int PH = H / 2, PW = W / 2;
   const int outs = COUT * PH * PW;
 
+  stage_to_lds(w, wlds, COUT * CIN * 9);
+
   for (int n = blockIdx.x; n < N; n += gridDim.x) {
-    stage_to_lds(in + (int64_t)n * plane, ilds, plane);
-    if (n == blockIdx.x) stage_to_lds(w, wlds, COUT * CIN * 9); // once per block
+    stage_plane_padded(in + (int64_t)n * plane, ilds, CIN, H, W);
     __syncthreads();
 
     for (int o = threadIdx.x; o < outs; o += kBlock) {
       const int px = o % PW;
       const int py = (o / PW) % PH;
       const int co = o / (PW * PH);
-      const int y0 = 2 * py, x0 = 2 * px;
+      const int y0 = 2 * py, x0 = 2 * px; // top-left of the padded 4x4 window
 
       float acc0 = bias[co], acc1 = acc0, acc2 = acc0, acc3 = acc0;
       const float* wc = wlds + co * CIN * 9;
       for (int ci = 0; ci < CIN; ++ci) {
-        const float* ip = ilds + ci * H * W;
+        const float* ip = ilds + ci * cs + y0 * PADW + x0;
         const float* wk = wc + ci * 9;
         float win[4][4];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const int yy = y0 - 1 + r;
 #pragma unroll
-          for (int c = 0; c < 4; ++c) {
-            const int xx = x0 - 1 + c;
-            win[r][c] = (yy >= 0 && yy < H && xx >= 0 && xx < W) ? ip[yy * W + xx] : 0.0f;
-          }
+          for (int c = 0; c < 4; ++c) win[r][c] = ip[r * PADW + c];
         }
 #pragma unroll
         for (int ky = 0; ky < 3; ++ky) {
@@ -113,38 +126,48 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_fwd_kernel(
 
 // ------------------------------------------------------------- bwd data
 
-// Workgroup = one sample. LDS: [COUT*H*W] dconv plane + [COUT*CIN*9] weights.
+// Scatter the relu+maxpool-folded conv gradient of one sample into a
+// halo-padded LDS plane (zeroed by the caller's stage).
+__device__ __forceinline__ void scatter_dconv_padded(const float* __restrict__ dpooled,
+                                                     const uint8_t* __restrict__ argmax,
+                                                     const float* __restrict__ pooled,
+                                                     float* __restrict__ dclds, int64_t n,
+                                                     int COUT, int H, int W) {
+  const int cs = padded_cstride(H, W), PADW = W + 2;
+  const int PH = H / 2, PW = W / 2;
+  const int cells = COUT * PH * PW;
+  for (int cell = threadIdx.x; cell < cells; cell += kBlock) {
+    const int64_t pidx = n * cells + cell;
+    const float pv = pooled[pidx];
+    if (pv <= 0.0f) continue;
+    const int sub = argmax[pidx];
+    const int px = cell % PW;
+    const int py = (cell / PW) % PH;
+    const int co = cell / (PW * PH);
+    const int yy = 2 * py + (sub >> 1);
+    const int xx = 2 * px + (sub & 1);
+    dclds[co * cs + (yy + 1) * PADW + (xx + 1)] = dpooled[pidx];
+  }
+}
+
+// Workgroup = one sample. LDS: [COUT][cstride] dconv plane + weights.
 __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_data_kernel(
     const float* __restrict__ dpooled, const uint8_t* __restrict__ argmax,
     const float* __restrict__ pooled, const float* __restrict__ w, float* __restrict__ din,
     int N, int CIN, int COUT, int H, int W) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* dclds = (float*)smem; // [COUT][H][W]
-  float* wlds = dclds + COUT * H * W; // [COUT][CIN][9]
+  const int cs = padded_cstride(H, W), PADW = W + 2;
+  float* dclds = (float*)smem; // [COUT][cs]
+  float* wlds = dclds + COUT * cs; // [COUT][CIN][9]
 
-  const int PH = H / 2, PW = W / 2;
-  const int cells = COUT * PH * PW;
-  const int dplane = COUT * H * W;
   const int iplane = CIN * H * W;
 
+  stage_to_lds(w, wlds, COUT * CIN * 9);
+
   for (int n = blockIdx.x; n < N; n += gridDim.x) {
-    if (n == blockIdx.x) stage_to_lds(w, wlds, COUT * CIN * 9);
-    // build the sparse dconv plane in LDS
-    for (int i = threadIdx.x; i < dplane; i += kBlock) dclds[i] = 0.0f;
+    for (int i = threadIdx.x; i < COUT * cs; i += kBlock) dclds[i] = 0.0f;
     __syncthreads();
-    for (int cell = threadIdx.x; cell < cells; cell += kBlock) {
-      const int64_t pidx = (int64_t)n * cells + cell;
-      const float pv = pooled[pidx];
-      if (pv <= 0.0f) continue;
-      const float g = dpooled[pidx];
-      const int sub = argmax[pidx];
-      const int px = cell % PW;
-      const int py = (cell / PW) % PH;
-      const int co = cell / (PW * PH);
-      const int yy = 2 * py + (sub >> 1);
-      const int xx = 2 * px + (sub & 1);
-      dclds[co * H * W + yy * W + xx] = g; // each cell owns its 2x2 patch
-    }
+    scatter_dconv_padded(dpooled, argmax, pooled, dclds, n, COUT, H, W);
     __syncthreads();
 
     for (int o = threadIdx.x; o < iplane; o += kBlock) {
@@ -153,17 +176,15 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_data_kernel(
       const int ci = o / (W * H);
       float acc = 0.0f;
       for (int co = 0; co < COUT; ++co) {
-        const float* dp = dclds + co * H * W;
+        // correlation with flipped kernel: dconv(y-ky+1, x-kx+1) ->
+        // padded row (y-ky+2), col (x-kx+2); halo absorbs the bounds.
+        const float* dp = dclds + co * cs + (y + 2) * PADW + (x + 2);
         const float* wk = wlds + (co * CIN + ci) * 9;
 #pragma unroll
         for (int ky = 0; ky < 3; ++ky) {
-          const int yy = y - ky + 1;
-          if (yy < 0 || yy >= H) continue;
 #pragma unroll
           for (int kx = 0; kx < 3; ++kx) {
-            const int xx = x - kx + 1;
-            if (xx < 0 || xx >= W) continue;
-            acc = fmaf(wk[ky * 3 + kx], dp[yy * W + xx], acc);
+            acc = fmaf(wk[ky * 3 + kx], dp[-ky * PADW - kx], acc);
           }
         }
       }
@@ -175,20 +196,22 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_data_kernel(
 
 // ----------------------------------------------------------- bwd weight
 
-// Workgroup = chunk of SAMPLES samples. LDS: inputs [S][CIN*H*W] +
-// dconv planes rebuilt sparse [S][COUT*H*W]. Thread owns (pair, slice):
-// pair = (co,ci), slices split the cell space; 9 register partials each,
-// LDS tree across slices, one global atomicAdd per (pair, tap).
+// Workgroup = a chunk of `samples` samples, looped one at a time through
+// LDS. Thread owns ((co,ci) pair, slice); its 9 register partials live
+// across the whole chunk; LDS tree across slices; one global atomicAdd
+// per tap per workgroup at the end.
 __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_weight_kernel(
     const float* __restrict__ dpooled, const uint8_t* __restrict__ argmax,
     const float* __restrict__ pooled, const float* __restrict__ in, float* __restrict__ dw,
     float* __restrict__ db, int N, int CIN, int COUT, int H, int W, int samples) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int cs = padded_cstride(H, W), PADW = W + 2;
   const int iplane = CIN * H * W;
-  const int dplane = COUT * H * W;
-  float* ilds = (float*)smem; // [S][CIN*H*W]
-  float* dclds = ilds + (int64_t)samples * iplane; // [S][COUT*H*W]
-  float* redlds = dclds + (int64_t)samples * dplane; // [kBlock] scratch
+  const int dplane = COUT * H * W; // dconv plane, unpadded (+4 bank skew)
+  const int dstride = H * W + 4;
+  float* ilds = (float*)smem; // [CIN][cs]
+  float* dclds = ilds + CIN * cs; // [COUT][dstride]
+  float* redlds = dclds + COUT * dstride; // [kBlock]
 
   const int PH = H / 2, PW = W / 2;
   const int cells = COUT * PH * PW;
@@ -204,61 +227,54 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_weight_kernel(
   const int n0 = blockIdx.x * samples;
   const int nvalid = min(samples, N - n0);
 
-  // stage inputs + rebuild dconv planes for the chunk
-  for (int s = 0; s < nvalid; ++s) {
-    stage_to_lds(in + (int64_t)(n0 + s) * iplane, ilds + (int64_t)s * iplane, iplane);
-  }
-  for (int64_t i = threadIdx.x; i < (int64_t)nvalid * dplane; i += kBlock) dclds[i] = 0.0f;
-  __syncthreads();
-  for (int64_t sc = threadIdx.x; sc < (int64_t)nvalid * cells; sc += kBlock) {
-    const int s = sc / cells;
-    const int cell = sc % cells;
-    const int64_t pidx = (int64_t)(n0 + s) * cells + cell;
-    const float pv = pooled[pidx];
-    if (pv <= 0.0f) continue;
-    const float g = dpooled[pidx];
-    const int sub = argmax[pidx];
-    const int px = cell % PW;
-    const int py = (cell / PW) % PH;
-    const int cco = cell / (PW * PH);
-    const int yy = 2 * py + (sub >> 1);
-    const int xx = 2 * px + (sub & 1);
-    dclds[(int64_t)s * dplane + cco * H * W + yy * W + xx] = g;
-  }
-  __syncthreads();
-
   float acc[9] = {0, 0, 0, 0, 0, 0, 0, 0, 0};
   float accb = 0.0f;
-  if (active) {
-    for (int s = 0; s < nvalid; ++s) {
-      const float* ip = ilds + (int64_t)s * iplane + ci * H * W;
-      const float* dp = dclds + (int64_t)s * dplane + co * H * W;
-      // iterate this co's conv positions, sliced across threads
+
+  for (int s = 0; s < nvalid; ++s) {
+    const int64_t n = n0 + s;
+    stage_plane_padded(in + n * iplane, ilds, CIN, H, W);
+    for (int i = threadIdx.x; i < COUT * dstride; i += kBlock) dclds[i] = 0.0f;
+    __syncthreads();
+    for (int cell = threadIdx.x; cell < cells; cell += kBlock) {
+      const int64_t pidx = n * cells + cell;
+      const float pv = pooled[pidx];
+      if (pv <= 0.0f) continue;
+      const int sub = argmax[pidx];
+      const int px = cell % PW;
+      const int py = (cell / PW) % PH;
+      const int cco = cell / (PW * PH);
+      const int yy = 2 * py + (sub >> 1);
+      const int xx = 2 * px + (sub & 1);
+      dclds[cco * dstride + yy * W + xx] = dpooled[pidx];
+    }
+    __syncthreads();
+
+    if (active) {
+      const float* ip = ilds + ci * cs;
+      const float* dp = dclds + co * dstride;
       for (int yx = slice; yx < H * W; yx += nslices) {
         const float g = dp[yx];
         if (g == 0.0f) continue;
-        const int yy = yx / W, xx = yx % W;
+        const int yy = yx / W, xx = yx - yy * W;
         if (ci == 0) accb += g;
+        // input window rows yy-1..yy+1 -> padded rows yy..yy+2
+        const float* iw = ip + yy * PADW + xx;
 #pragma unroll
         for (int ky = 0; ky < 3; ++ky) {
-          const int iy = yy + ky - 1;
-          if (iy < 0 || iy >= H) continue;
 #pragma unroll
           for (int kx = 0; kx < 3; ++kx) {
-            const int ix = xx + kx - 1;
-            if (ix < 0 || ix >= W) continue;
-            acc[ky * 3 + kx] = fmaf(g, ip[iy * W + ix], acc[ky * 3 + kx]);
+            acc[ky * 3 + kx] = fmaf(g, iw[ky * PADW + kx], acc[ky * 3 + kx]);
           }
         }
       }
     }
+    __syncthreads();
   }
 
   // cross-slice reduction (skipped when nslices == 1)
   if (nslices > 1) {
 #pragma unroll
     for (int k = 0; k < 9; ++k) {
-      __syncthreads();
       redlds[threadIdx.x] = active ? acc[k] : 0.0f;
       __syncthreads();
       if (active && slice == 0) {
@@ -266,8 +282,8 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_weight_kernel(
         for (int sl = 1; sl < nslices; ++sl) total += redlds[sl * pairs + pair];
         acc[k] = total;
       }
+      __syncthreads();
     }
-    __syncthreads();
     redlds[threadIdx.x] = (active && ci == 0) ? accb : 0.0f;
     __syncthreads();
     if (active && slice == 0 && ci == 0) {
@@ -297,7 +313,7 @@ void conv3x3_relu_pool_fwd(at::Tensor in, at::Tensor w, at::Tensor b, at::Tensor
   const int N = in.size(0), CIN = in.size(1), H = in.size(2), W = in.size(3);
   const int COUT = w.size(0);
   TORCH_CHECK(H % 2 == 0 && W % 2 == 0, "H and W must be even for 2x2 pooling");
-  const int lds_bytes = (CIN * H * W + COUT * CIN * 9) * (int)sizeof(float);
+  const int lds_bytes = (CIN * padded_cstride(H, W) + COUT * CIN * 9) * (int)sizeof(float);
   TORCH_CHECK(lds_bytes <= kMaxLds, "plane+weights exceed LDS (", lds_bytes, " B)");
   auto stream = c10::hip::getCurrentHIPStream();
   const int blocks = std::min(N, kMaxGrid);
@@ -310,7 +326,7 @@ void conv3x3_relu_pool_bwd_data(at::Tensor dpooled, at::Tensor argmax, at::Tenso
                                 at::Tensor w, at::Tensor din) {
   const int N = din.size(0), CIN = din.size(1), H = din.size(2), W = din.size(3);
   const int COUT = w.size(0);
-  const int lds_bytes = (COUT * H * W + COUT * CIN * 9) * (int)sizeof(float);
+  const int lds_bytes = (COUT * padded_cstride(H, W) + COUT * CIN * 9) * (int)sizeof(float);
   TORCH_CHECK(lds_bytes <= kMaxLds, "dconv plane exceeds LDS");
   auto stream = c10::hip::getCurrentHIPStream();
   const int blocks = std::min(N, kMaxGrid);
@@ -325,17 +341,13 @@ void conv3x3_relu_pool_bwd_weight(at::Tensor dpooled, at::Tensor argmax, at::Ten
   const int N = in.size(0), CIN = in.size(1), H = in.size(2), W = in.size(3);
   const int COUT = dw.size(0);
   TORCH_CHECK(COUT * CIN <= kBlock, "bwd_weight supports COUT*CIN <= ", kBlock);
-  const int iplane = CIN * H * W, dplane = COUT * H * W;
-  // pick the largest chunk that fits a 64 KiB LDS budget (>= 2 blocks/CU
-  // for latency hiding; kBlock floats of reduction scratch included)
-  constexpr int kWeightLdsBudget = 64 * 1024;
+  const int lds_bytes =
+      (CIN * padded_cstride(H, W) + COUT * (H * W + 4) + kBlock) * (int)sizeof(float);
+  TORCH_CHECK(lds_bytes <= kMaxLds, "bwd_weight staging exceeds LDS (", lds_bytes, " B)");
+  // chunk so that the grid stays ~>= 512 blocks (occupancy) while cutting
+  // the per-chunk global atomics by the chunk factor
   int samples = 1;
-  while (samples < 16 &&
-         ((int64_t)(samples * 2) * (iplane + dplane) + kBlock) * (int64_t)sizeof(float) <=
-             kWeightLdsBudget)
-    samples *= 2;
-  const int lds_bytes = (samples * (iplane + dplane) + kBlock) * (int)sizeof(float);
-  TORCH_CHECK(lds_bytes <= kMaxLds, "bwd_weight staging exceeds LDS");
+  while (samples < 16 && (N + samples * 2 - 1) / (samples * 2) >= 512) samples *= 2;
   const int blocks = (N + samples - 1) / samples;
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(conv3x3_relu_pool_bwd_weight_kernel, dim3(blocks), dim3(kBlock), lds_bytes,
