@@ -5,7 +5,10 @@ Prints per-kernel times (cuda events, 100 reps) for the two MNIST-CNN
 layer shapes, plus the torch/MIOpen equivalents for comparison.
 """
 
+import os
 import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
