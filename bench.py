@@ -29,6 +29,8 @@ import os
 import sys
 import time
 
+os.environ.setdefault('HSA_ENABLE_IPC_MODE_LEGACY', '0')  # dmabuf IPC for RCCL
+
 import torch
 import torch.distributed as dist
 
@@ -190,7 +192,7 @@ def main():
         device = torch.device('cpu')
 
     if args.batch_size is None:
-        args.batch_size = {'mnist': 1024, 'resnet50': 256, 'gpt2': 8}[args.model]
+        args.batch_size = {'mnist': 16384, 'resnet50': 256, 'gpt2': 16}[args.model]
 
     # Stage + pipeline machinery (the metric path under test runs per step)
     pipeline = TrainingPipeline(name='bench')

@@ -26,6 +26,7 @@ from .parallel.distributed import is_root
 from .parallel.flat import FlatOptimizer
 from .utils.logging import DevNullIO, flush_log_handlers
 from .utils.table import ProgressTable
+from .utils.tracing import roctx_range
 
 __all__ = ['Stage', 'TrainValStage']
 
@@ -119,7 +120,8 @@ class Stage:
         self._pre_stage()
         while self.max_epochs is None or self.current_epoch <= self.max_epochs:
             self._pre_epoch()
-            self.run_epoch()
+            with roctx_range(f'epoch_{self.current_epoch}'):
+                self.run_epoch()
             self._post_epoch()
             if self._stop_requested:
                 break
@@ -290,9 +292,10 @@ class TrainValStage(Stage):
         (grad sync) -> optimizer, plus the standard step metrics. The
         benchmark driver calls this directly."""
         step_start_time = time.perf_counter_ns()
-        self.zero_grad()
-        loss = self.train_step(batch)
-        self.optimize(loss)
+        with roctx_range('train_batch'):
+            self.zero_grad()
+            loss = self.train_step(batch)
+            self.optimize(loss)
         step_end_time = time.perf_counter_ns()
 
         self.track_reduce(self.loss_metric_name(), loss)
