@@ -56,7 +56,7 @@ def parse_args():
     p.add_argument('--batch-size', type=int, default=None, help='per-GPU batch size')
     p.add_argument('--no-graph', action='store_true', help='disable hipGraph capture')
     p.add_argument('--no-fused', action='store_true', help='disable the fused smallcnn kernels (MIOpen path)')
-    p.add_argument('--channels-last', action='store_true', help='NHWC memory format (resnet50)')
+    p.add_argument('--no-channels-last', action='store_true', help='disable NHWC memory format (resnet50)')
     p.add_argument('--seq-len', type=int, default=1024, help='gpt2 sequence length')
     return p.parse_args()
 
@@ -90,6 +90,7 @@ class BenchStage(TrainValStage):
             self.dtype = 'fp32'
         elif args.model == 'resnet50':
             model = resnet50()
+            args.channels_last = not args.no_channels_last and device.type == 'cuda'
             if args.channels_last:
                 model = model.to(memory_format=torch.channels_last)
                 torch.backends.cudnn.benchmark = True
