@@ -52,7 +52,12 @@ __host__ __device__ __forceinline__ PlaneGeom plane_geom(int H, int W) {
   int padw = W + 2 + ((PW & 1) ? 2 : 0);
   g.PADW = (padw + 3) & ~3;
   g.cs = g.PADH * g.PADW;
-  if (g.cs % 64 == 0) g.cs += 4;
+  // keep cs a multiple of 4 (16B-aligned rows) with cs/4 odd, so the
+  // per-channel bank offset (cs mod 64) has gcd(cs,64)=4 and 16
+  // simultaneous cross-channel b32 reads at equal (y,x) span 16 distinct
+  // banks (measured: cs=360 aliased ci and ci+8 -> 5.2 conflict
+  // cycles/LDS instruction in bwd_weight, profiles/smallcnn_pmc.txt)
+  if ((g.cs & 7) == 0) g.cs += 4; // cs is a multiple of 4 by construction
   return g;
 }
 
