@@ -101,8 +101,12 @@ class BenchStage(TrainValStage):
             self.batch_shape = (args.batch_size, args.seq_len)
             self.dtype = 'bf16'
 
+        # bf16 configs run as true mixed precision: bf16 flat params/grads,
+        # fp32 master in the fused optimizer — no autocast casting per layer
+        self.flat_bf16 = args.impl == 'flat' and self.dtype == 'bf16' and device.type == 'cuda'
         if args.impl == 'flat':
-            self.pipeline.register_model('net', model, ddp_impl='flat', verbose=False)
+            flat_dtype = torch.bfloat16 if self.flat_bf16 else torch.float32
+            self.pipeline.register_model('net', model, ddp_impl='flat', flat_dtype=flat_dtype, verbose=False)
             replica = self.pipeline.models['net']
             if args.model == 'resnet50':
                 self.pipeline.register_optimizer('opt', FlatSGD(replica, lr=1e-3, momentum=0.9))
@@ -128,8 +132,9 @@ class BenchStage(TrainValStage):
             self.static_batch = torch.zeros_like(self.pool[0])
         else:
             mf = torch.channels_last if (args.model == 'resnet50' and args.channels_last) else torch.contiguous_format
+            in_dtype = torch.bfloat16 if (args.model == 'resnet50' and self.flat_bf16) else torch.float32
             self.pool = [
-                torch.randn(self.batch_shape, generator=g).to(device).to(memory_format=mf)
+                torch.randn(self.batch_shape, generator=g).to(device, in_dtype).to(memory_format=mf)
                 for _ in range(self.n_pool)
             ]
             self.labels = [
@@ -143,14 +148,17 @@ class BenchStage(TrainValStage):
 
     def step(self, batch):
         model = self.pipeline.models['net']
+        autocast_on = (
+            self.dtype == 'bf16' and self.bench_device.type == 'cuda' and not self.flat_bf16
+        )  # bf16 flat replicas compute natively in bf16 — no autocast needed
         if self.args.model == 'gpt2':
             idx = batch
-            with torch.autocast('cuda', dtype=torch.bfloat16, enabled=self.bench_device.type == 'cuda'):
+            with torch.autocast('cuda', dtype=torch.bfloat16, enabled=autocast_on):
                 _, loss = model(idx, targets=idx)
             return loss
         x, y = batch
         if self.args.model == 'resnet50':
-            with torch.autocast('cuda', dtype=torch.bfloat16, enabled=self.bench_device.type == 'cuda'):
+            with torch.autocast('cuda', dtype=torch.bfloat16, enabled=autocast_on):
                 out = model(x)
                 loss = self.loss_fn(out, y)
             return loss

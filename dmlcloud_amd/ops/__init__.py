@@ -132,6 +132,35 @@ def fused_adam(param, grad, exp_avg, exp_avg_sq, step_t, lr, beta1, beta2, eps, 
         ref.fused_adam_step(param, grad, exp_avg, exp_avg_sq, step_t, lr, beta1, beta2, eps, weight_decay, grad_scale)
 
 
+def fused_adam_bf16(param, grad, master, exp_avg, exp_avg_sq, step_t, lr, beta1, beta2, eps, weight_decay, grad_scale=1.0):
+    if param.is_cuda:
+        _require_ext()
+        _C.fused_adam_bf16(param, grad, master, exp_avg, exp_avg_sq, step_t, lr, beta1, beta2, eps, weight_decay, grad_scale)
+    else:
+        ref.fused_adam_bf16_step(
+            param, grad, master, exp_avg, exp_avg_sq, step_t, lr, beta1, beta2, eps, weight_decay, grad_scale
+        )
+
+
+def fused_sgd_bf16(param, grad, master, momentum_buf, lr, momentum, weight_decay, grad_scale=1.0):
+    use_momentum = momentum_buf is not None and momentum != 0
+    if param.is_cuda:
+        _require_ext()
+        _C.fused_sgd_bf16(
+            param,
+            grad,
+            master,
+            momentum_buf if momentum_buf is not None else torch.Tensor(),
+            lr,
+            momentum,
+            weight_decay,
+            grad_scale,
+            use_momentum,
+        )
+    else:
+        ref.fused_sgd_bf16_step(param, grad, master, momentum_buf if use_momentum else None, lr, momentum, weight_decay, grad_scale)
+
+
 def fused_sgd(param, grad, momentum_buf, lr, momentum, weight_decay, grad_scale=1.0):
     use_momentum = momentum_buf is not None and momentum != 0
     if param.is_cuda:

@@ -106,6 +106,55 @@ def fused_adam_step(
     param.addcdiv_(exp_avg / bc1, denom, value=-lr)
 
 
+def fused_adam_bf16_step(
+    param: torch.Tensor,
+    grad: torch.Tensor,
+    master: torch.Tensor,
+    exp_avg: torch.Tensor,
+    exp_avg_sq: torch.Tensor,
+    step_t: torch.Tensor,
+    lr: float,
+    beta1: float,
+    beta2: float,
+    eps: float,
+    weight_decay: float,
+    grad_scale: float,
+):
+    """Mixed-precision Adam: bf16 params/grads, fp32 master + moments."""
+    step_t += 1
+    t = step_t.to(torch.float32)
+    bc1 = 1 - beta1**t
+    bc2 = 1 - beta2**t
+    g = grad.to(torch.float32) * grad_scale
+    if weight_decay != 0:
+        g = g + weight_decay * master
+    exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
+    exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    denom = (exp_avg_sq / bc2).sqrt_().add_(eps)
+    master.addcdiv_(exp_avg / bc1, denom, value=-lr)
+    param.copy_(master.to(torch.bfloat16))
+
+
+def fused_sgd_bf16_step(
+    param: torch.Tensor,
+    grad: torch.Tensor,
+    master: torch.Tensor,
+    momentum_buf: Optional[torch.Tensor],
+    lr: float,
+    momentum: float,
+    weight_decay: float,
+    grad_scale: float,
+):
+    g = grad.to(torch.float32) * grad_scale
+    if weight_decay != 0:
+        g = g + weight_decay * master
+    if momentum_buf is not None and momentum != 0:
+        momentum_buf.mul_(momentum).add_(g)
+        g = momentum_buf
+    master.add_(g, alpha=-lr)
+    param.copy_(master.to(torch.bfloat16))
+
+
 def fused_sgd_step(
     param: torch.Tensor,
     grad: torch.Tensor,

@@ -23,6 +23,12 @@ void fused_adam(at::Tensor param, at::Tensor grad, at::Tensor exp_avg, at::Tenso
                 double weight_decay, double grad_scale);
 void fused_sgd(at::Tensor param, at::Tensor grad, at::Tensor momentum_buf, double lr,
                double momentum, double weight_decay, double grad_scale, bool use_momentum);
+void fused_adam_bf16(at::Tensor param, at::Tensor grad, at::Tensor master, at::Tensor exp_avg,
+                     at::Tensor exp_avg_sq, at::Tensor step_t, double lr, double beta1,
+                     double beta2, double eps, double weight_decay, double grad_scale);
+void fused_sgd_bf16(at::Tensor param, at::Tensor grad, at::Tensor master,
+                    at::Tensor momentum_buf, double lr, double momentum, double weight_decay,
+                    double grad_scale, bool use_momentum);
 void l2_norm_and_scale(at::Tensor flat, at::Tensor partials, at::Tensor out, double max_norm,
                        bool apply);
 
@@ -47,6 +53,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("chunked_copy", &dmlamd::chunked_copy, "Descriptor-table gather/scatter copy");
   m.def("fused_adam", &dmlamd::fused_adam, "Fused Adam on flat fp32 buffers");
   m.def("fused_sgd", &dmlamd::fused_sgd, "Fused SGD on flat fp32 buffers");
+  m.def("fused_adam_bf16", &dmlamd::fused_adam_bf16,
+        "Fused Adam: bf16 params/grads, fp32 master + moments");
+  m.def("fused_sgd_bf16", &dmlamd::fused_sgd_bf16,
+        "Fused SGD: bf16 params/grads, fp32 master");
   m.def("l2_norm_and_scale", &dmlamd::l2_norm_and_scale,
         "Deterministic L2 norm + optional clip scale");
   m.def("conv3x3_relu_pool_fwd", &dmlamd::conv3x3_relu_pool_fwd,

@@ -30,27 +30,33 @@ import torch.distributed as dist
 from .. import ops
 
 
-def _flatten_params(params: List[torch.nn.Parameter], device=None):
-    """Move `params` into one flat fp32 buffer; returns (flat, grads_flat)."""
+def _flatten_params(params: List[torch.nn.Parameter], dtype=torch.float32, device=None):
+    """Move `params` into one flat buffer of `dtype`; returns
+    (flat, grads_flat, master) where master is the fp32 copy (None when
+    dtype is already fp32)."""
     assert params, 'no trainable parameters'
     device = device or params[0].device
     numels = [p.numel() for p in params]
-    # 4-element (16 B) alignment per param so every view is vector-load friendly
+    # 8-element (16 B for bf16) alignment so every view is vector-load friendly
     offsets = []
     off = 0
     for n in numels:
         offsets.append(off)
-        off += (n + 3) // 4 * 4
+        off += (n + 7) // 8 * 8
     total = off
 
-    flat = torch.zeros(total, dtype=torch.float32, device=device)
-    grad_flat = torch.zeros(total, dtype=torch.float32, device=device)
+    flat = torch.zeros(total, dtype=dtype, device=device)
+    grad_flat = torch.zeros(total, dtype=dtype, device=device)
+    master = torch.zeros(total, dtype=torch.float32, device=device) if dtype != torch.float32 else None
 
     for p, o, n in zip(params, offsets, numels):
-        flat[o : o + n].copy_(p.data.reshape(-1).to(device=device, dtype=torch.float32))
+        fp32 = p.data.reshape(-1).to(device=device, dtype=torch.float32)
+        if master is not None:
+            master[o : o + n].copy_(fp32)
+        flat[o : o + n].copy_(fp32.to(dtype))
         p.data = flat[o : o + n].view(p.shape)
         p.grad = grad_flat[o : o + n].view(p.shape)
-    return flat, grad_flat
+    return flat, grad_flat, master
 
 
 class FlatReplica:
@@ -65,14 +71,27 @@ class FlatReplica:
         optimizer.step()                       # FlatAdam/FlatSGD
     """
 
-    def __init__(self, module: torch.nn.Module, process_group=None, broadcast: bool = True):
+    def __init__(
+        self,
+        module: torch.nn.Module,
+        process_group=None,
+        broadcast: bool = True,
+        dtype: torch.dtype = torch.float32,
+    ):
+        """dtype=torch.bfloat16 enables true mixed precision: the model
+        computes in bf16 (no per-layer autocast casts), gradients
+        all-reduce in bf16 (half the xGMI bytes), and the fused optimizer
+        holds the fp32 master (replica.flat_master)."""
+        if dtype not in (torch.float32, torch.bfloat16):
+            raise ValueError('FlatReplica supports fp32 or bf16 parameter buffers')
         self.module = module
+        self.dtype = dtype
         self.process_group = process_group
         self.params = [p for p in module.parameters() if p.requires_grad]
         for p in self.params:
             if p.dtype != torch.float32:
-                raise ValueError('FlatReplica requires fp32 parameters (use torch DDP for mixed dtypes)')
-        self.flat_param, self.flat_grad = _flatten_params(self.params)
+                raise ValueError('FlatReplica expects fp32 module parameters at construction')
+        self.flat_param, self.flat_grad, self.flat_master = _flatten_params(self.params, dtype=dtype)
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         if broadcast and dist.is_initialized() and self.world_size > 1:
             dist.broadcast(self.flat_param, src=0, group=process_group)
@@ -146,19 +165,35 @@ class FlatAdam(FlatOptimizer):
         self.param_groups[0]['lr'] = lr
 
     def step(self):
-        ops.fused_adam(
-            self.replica.flat_param,
-            self.replica.flat_grad,
-            self.exp_avg,
-            self.exp_avg_sq,
-            self.step_t,
-            self.lr,
-            self.beta1,
-            self.beta2,
-            self.eps,
-            self.weight_decay,
-            grad_scale=self.replica.grad_scale,
-        )
+        if self.replica.dtype is torch.bfloat16:
+            ops.fused_adam_bf16(
+                self.replica.flat_param,
+                self.replica.flat_grad,
+                self.replica.flat_master,
+                self.exp_avg,
+                self.exp_avg_sq,
+                self.step_t,
+                self.lr,
+                self.beta1,
+                self.beta2,
+                self.eps,
+                self.weight_decay,
+                grad_scale=self.replica.grad_scale,
+            )
+        else:
+            ops.fused_adam(
+                self.replica.flat_param,
+                self.replica.flat_grad,
+                self.exp_avg,
+                self.exp_avg_sq,
+                self.step_t,
+                self.lr,
+                self.beta1,
+                self.beta2,
+                self.eps,
+                self.weight_decay,
+                grad_scale=self.replica.grad_scale,
+            )
 
     def state_dict(self):
         return {
@@ -169,6 +204,7 @@ class FlatAdam(FlatOptimizer):
             'exp_avg': self.exp_avg,
             'exp_avg_sq': self.exp_avg_sq,
             'step': self.step_t,
+            'master': self.replica.flat_master,
         }
 
     def load_state_dict(self, state):
@@ -179,6 +215,8 @@ class FlatAdam(FlatOptimizer):
         self.exp_avg.copy_(state['exp_avg'].to(self.exp_avg.device))
         self.exp_avg_sq.copy_(state['exp_avg_sq'].to(self.exp_avg_sq.device))
         self.step_t.copy_(state['step'].to(self.step_t.device))
+        if state.get('master') is not None and self.replica.flat_master is not None:
+            self.replica.flat_master.copy_(state['master'].to(self.replica.flat_master.device))
 
 
 class FlatSGD(FlatOptimizer):
@@ -189,19 +227,34 @@ class FlatSGD(FlatOptimizer):
         self.weight_decay = weight_decay
         self.momentum_buf: Optional[torch.Tensor] = None
         if momentum != 0:
-            self.momentum_buf = torch.zeros_like(replica.flat_param)
+            # momentum accumulates in fp32 regardless of the param dtype
+            self.momentum_buf = torch.zeros(
+                replica.flat_param.numel(), dtype=torch.float32, device=replica.flat_param.device
+            )
         self.param_groups[0]['lr'] = lr
 
     def step(self):
-        ops.fused_sgd(
-            self.replica.flat_param,
-            self.replica.flat_grad,
-            self.momentum_buf,
-            self.lr,
-            self.momentum,
-            self.weight_decay,
-            grad_scale=self.replica.grad_scale,
-        )
+        if self.replica.dtype is torch.bfloat16:
+            ops.fused_sgd_bf16(
+                self.replica.flat_param,
+                self.replica.flat_grad,
+                self.replica.flat_master,
+                self.momentum_buf,
+                self.lr,
+                self.momentum,
+                self.weight_decay,
+                grad_scale=self.replica.grad_scale,
+            )
+        else:
+            ops.fused_sgd(
+                self.replica.flat_param,
+                self.replica.flat_grad,
+                self.momentum_buf,
+                self.lr,
+                self.momentum,
+                self.weight_decay,
+                grad_scale=self.replica.grad_scale,
+            )
 
     def state_dict(self):
         return {
@@ -209,6 +262,7 @@ class FlatSGD(FlatOptimizer):
             'momentum': self.momentum,
             'weight_decay': self.weight_decay,
             'momentum_buf': self.momentum_buf,
+            'master': self.replica.flat_master,
         }
 
     def load_state_dict(self, state):
@@ -217,5 +271,11 @@ class FlatSGD(FlatOptimizer):
         self.weight_decay = state['weight_decay']
         if state['momentum_buf'] is not None:
             if self.momentum_buf is None:
-                self.momentum_buf = torch.zeros_like(self.replica.flat_param)
+                self.momentum_buf = torch.zeros(
+                    self.replica.flat_param.numel(),
+                    dtype=torch.float32,
+                    device=self.replica.flat_param.device,
+                )
             self.momentum_buf.copy_(state['momentum_buf'].to(self.momentum_buf.device))
+        if state.get('master') is not None and self.replica.flat_master is not None:
+            self.replica.flat_master.copy_(state['master'].to(self.replica.flat_master.device))

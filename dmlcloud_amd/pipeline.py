@@ -89,6 +89,7 @@ class TrainingPipeline:
         use_ddp: bool = True,
         sync_bn: bool = False,
         ddp_impl: str = 'torch',
+        flat_dtype: torch.dtype = torch.float32,
         bucket_cap_mb: Optional[int] = None,
         save_latest: bool = True,
         save_interval: Optional[int] = None,
@@ -112,7 +113,7 @@ class TrainingPipeline:
             model = model.to(self.device)
             if sync_bn:
                 model = torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
-            model = FlatReplica(model)
+            model = FlatReplica(model, dtype=flat_dtype)
         else:
             model = model.to(self.device)
             if dist.is_initialized() and dist.get_world_size() > 1:

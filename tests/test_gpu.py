@@ -170,6 +170,37 @@ class TestOptimKernels:
             ops.fused_sgd(param_dev, grad.to(DEV), mom_dev, 0.1, 0.9, 0.001, 1.0)
         torch.testing.assert_close(param_dev.cpu(), param_ref, rtol=1e-5, atol=1e-6)
 
+    def test_adam_bf16_matches_reference(self):
+        _assert_native_loaded()
+        torch.manual_seed(0)
+        n = (1 << 16) + 5
+        master_ref = torch.randn(n)
+        param_ref = master_ref.to(torch.bfloat16)
+        m_ref, v_ref = torch.zeros(n), torch.zeros(n)
+        st_ref = torch.zeros(1, dtype=torch.int32)
+
+        master_dev = master_ref.to(DEV)
+        param_dev = param_ref.to(DEV)
+        m_dev, v_dev = torch.zeros(n, device=DEV), torch.zeros(n, device=DEV)
+        st_dev = torch.zeros(1, dtype=torch.int32, device=DEV)
+
+        for it in range(3):
+            torch.manual_seed(it)
+            grad = torch.randn(n).to(torch.bfloat16)
+            ref.fused_adam_bf16_step(param_ref, grad, master_ref, m_ref, v_ref, st_ref, 1e-2, 0.9, 0.999, 1e-8, 0.01, 0.5)
+            ops.fused_adam_bf16(param_dev, grad.to(DEV), master_dev, m_dev, v_dev, st_dev, 1e-2, 0.9, 0.999, 1e-8, 0.01, 0.5)
+        torch.testing.assert_close(master_dev.cpu(), master_ref, rtol=1e-4, atol=1e-6)
+        torch.testing.assert_close(param_dev.cpu(), param_ref)
+
+    def test_clip_bf16(self):
+        _assert_native_loaded()
+        g = torch.randn(1 << 20, device=DEV).to(torch.bfloat16)
+        g_ref = g.float().cpu()
+        norm = ops.clip_grad_norm_(g, 0.5)
+        expected_norm = g_ref.norm()
+        assert norm.item() == pytest.approx(expected_norm.item(), rel=1e-2)
+        assert g.float().norm().item() == pytest.approx(0.5, rel=1e-2)
+
     def test_clip_matches_torch(self):
         _assert_native_loaded()
         torch.manual_seed(0)

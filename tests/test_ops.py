@@ -64,6 +64,54 @@ class TestFusedAdamReference:
         torch.testing.assert_close(param_a, param_b)
 
 
+class TestFusedAdamBf16Reference:
+    def test_master_tracks_fp32_adam(self):
+        """bf16 Adam's fp32 master must follow a plain fp32 Adam driven by
+        the (bf16-rounded) gradients."""
+        torch.manual_seed(0)
+        n = 511
+        master = torch.randn(n)
+        param = master.to(torch.bfloat16)
+        m1, v1 = torch.zeros(n), torch.zeros(n)
+        st1 = torch.zeros(1, dtype=torch.int32)
+
+        ref_p = master.clone()
+        m2, v2 = torch.zeros(n), torch.zeros(n)
+        st2 = torch.zeros(1, dtype=torch.int32)
+
+        for it in range(3):
+            torch.manual_seed(it)
+            grad32 = torch.randn(n)
+            grad16 = grad32.to(torch.bfloat16)
+            ops.fused_adam_bf16(param, grad16, master, m1, v1, st1, 1e-2, 0.9, 0.999, 1e-8, 0.0)
+            ops.fused_adam(ref_p, grad16.to(torch.float32), m2, v2, st2, 1e-2, 0.9, 0.999, 1e-8, 0.0)
+        torch.testing.assert_close(master, ref_p, rtol=1e-6, atol=1e-7)
+        torch.testing.assert_close(param, master.to(torch.bfloat16))
+
+
+class TestFlatReplicaBf16:
+    def test_bf16_views_and_master(self):
+        from dmlcloud_amd.parallel import FlatAdam, FlatReplica
+
+        torch.manual_seed(0)
+        model = torch.nn.Linear(8, 4)
+        w0 = model.weight.detach().clone()
+        replica = FlatReplica(model, broadcast=False, dtype=torch.bfloat16)
+        assert replica.flat_param.dtype == torch.bfloat16
+        assert replica.flat_master.dtype == torch.float32
+        assert model.weight.dtype == torch.bfloat16
+        torch.testing.assert_close(model.weight.detach().float(), w0, rtol=1e-2, atol=1e-2)
+
+        opt = FlatAdam(replica, lr=1e-2)
+        x = torch.randn(4, 8, dtype=torch.bfloat16)
+        replica.zero_grad()
+        replica(x).float().pow(2).mean().backward()
+        assert model.weight.grad.dtype == torch.bfloat16
+        opt.step()
+        # params and master moved together
+        torch.testing.assert_close(replica.flat_param.float(), replica.flat_master, rtol=1e-2, atol=1e-2)
+
+
 class TestFusedSgdReference:
     def test_matches_torch_sgd_momentum(self):
         torch.manual_seed(0)
