@@ -1,0 +1,142 @@
+"""Pipeline checkpoint round-trip, graphs fallback, tracing, misc."""
+
+import sys
+
+import pytest
+import torch
+
+from dmlcloud_amd import TrainingPipeline, TrainValStage
+
+
+class DS(torch.utils.data.Dataset):
+    def __len__(self):
+        return 8
+
+    def __getitem__(self, idx):
+        g = torch.Generator().manual_seed(idx)
+        return torch.randn(10, generator=g), idx % 10
+
+
+class Stage_(TrainValStage):
+    def pre_stage(self):
+        torch.manual_seed(0)
+        model = torch.nn.Linear(10, 10)
+        self.pipeline.register_model('m', model, save_interval=1)
+        self.pipeline.register_optimizer('sgd', torch.optim.SGD(model.parameters(), lr=1e-2))
+        self.pipeline.register_dataset('train', torch.utils.data.DataLoader(DS(), batch_size=4))
+        self.pipeline.register_dataset('val', torch.utils.data.DataLoader(DS(), batch_size=4))
+        self.loss = torch.nn.CrossEntropyLoss()
+
+    def step(self, batch):
+        x, y = batch
+        return self.loss(self.pipeline.models['m'](x), y)
+
+
+class TestCheckpointIntegration:
+    def test_model_files_written(self, torch_distributed, tmp_path):
+        pipeline = TrainingPipeline(name='ckpt')
+        pipeline.enable_checkpointing(str(tmp_path), resume=False)
+        pipeline.append_stage(Stage_(), max_epochs=2)
+        pipeline.run()
+
+        ckpt = pipeline.checkpoint_dir
+        assert ckpt.is_valid
+        assert ckpt.config_file.exists()
+        model_dir = ckpt.models_dir / 'm'
+        assert (model_dir / 'latest.dmlt').exists()
+        assert (model_dir / 'epoch_0001.dmlt').exists()
+        assert (model_dir / 'epoch_0002.dmlt').exists()
+
+    def test_full_state_roundtrip(self, torch_distributed, tmp_path):
+        pipeline = TrainingPipeline(name='ckpt')
+        pipeline.enable_checkpointing(str(tmp_path), resume=False)
+        stage = Stage_()
+        pipeline.append_stage(stage, max_epochs=2)
+        pipeline.run()
+        pipeline.save_checkpoint()
+        weight = pipeline.models['m'].module.weight.detach().clone()
+        loss_history = pipeline.tracker['train/loss']
+
+        # fresh pipeline resumes from the saved state
+        pipeline2 = TrainingPipeline(name='ckpt2')
+        pipeline2.checkpoint_dir = pipeline.checkpoint_dir
+        pipeline2.device = torch.device('cpu')
+        stage2 = Stage_()
+        pipeline2.append_stage(stage2, max_epochs=2)
+        stage2.pre_stage()
+        pipeline2.load_checkpoint()
+        torch.testing.assert_close(pipeline2.models['m'].module.weight.detach(), weight)
+        assert len(pipeline2.tracker['train/loss']) == len(loss_history)
+        assert stage2.current_epoch == stage.current_epoch
+
+    def test_resume_flag(self, torch_distributed, tmp_path):
+        p1 = TrainingPipeline(name='r')
+        p1.enable_checkpointing(str(tmp_path), resume=False)
+        p1.append_stage(Stage_(), max_epochs=1)
+        p1.run()
+        path = str(p1.checkpoint_dir.path)
+
+        p2 = TrainingPipeline(name='r')
+        p2.enable_checkpointing(path, resume=True)
+        assert p2.resumed is True
+
+        p3 = TrainingPipeline(name='r')
+        p3.enable_checkpointing(str(tmp_path / 'other'), resume=True)
+        assert p3.resumed is False
+
+
+class TestGraphedStepCpu:
+    def test_eager_fallback(self):
+        from dmlcloud_amd.parallel import GraphedStep
+
+        calls = []
+        gs = GraphedStep(lambda: calls.append(1), enabled=False)
+        gs.initialize()
+        assert not gs.captured
+        gs()
+        gs()
+        assert len(calls) == 2
+
+
+class TestTracing:
+    def test_roctx_noop_cpu(self):
+        from dmlcloud_amd.utils.tracing import enable_tracing, roctx_range
+
+        enable_tracing(True)
+        try:
+            with roctx_range('x'):
+                pass
+        finally:
+            enable_tracing(False)
+        with roctx_range('y'):
+            pass
+
+
+class TestFlatBf16Cpu:
+    def test_bf16_pipeline_flat(self, torch_distributed):
+        from dmlcloud_amd.parallel import FlatAdam
+
+        class FlatStage(Stage_):
+            def pre_stage(self):
+                torch.manual_seed(0)
+                model = torch.nn.Linear(10, 10)
+                self.pipeline.register_model('m', model, ddp_impl='flat', flat_dtype=torch.bfloat16)
+                replica = self.pipeline.models['m']
+                self.pipeline.register_optimizer('adam', FlatAdam(replica, lr=1e-3))
+                self.pipeline.register_dataset('train', torch.utils.data.DataLoader(DS(), batch_size=4))
+                self.pipeline.register_dataset('val', torch.utils.data.DataLoader(DS(), batch_size=4))
+                self.loss = torch.nn.CrossEntropyLoss()
+
+            def step(self, batch):
+                x, y = batch
+                out = self.pipeline.models['m'](x.to(torch.bfloat16))
+                return self.loss(out.float(), y)
+
+        pipeline = TrainingPipeline()
+        pipeline.append_stage(FlatStage(), max_epochs=1)
+        pipeline.run()
+        assert pipeline.tracker['train/loss'][0] is not None
+
+
+if __name__ == '__main__':
+    sys.exit(pytest.main([__file__]))
