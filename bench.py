@@ -206,7 +206,7 @@ def main():
         device = torch.device('cpu')
 
     if args.batch_size is None:
-        args.batch_size = {'mnist': 16384, 'resnet50': 256, 'gpt2': 64}[args.model]
+        args.batch_size = {'mnist': 16384, 'resnet50': 512, 'gpt2': 64}[args.model]
 
     # Stage + pipeline machinery (the metric path under test runs per step)
     pipeline = TrainingPipeline(name='bench')

@@ -330,8 +330,10 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_weight_kernel(
       const float* gp = glds + co * pstride;
       const float* sp = slds + co * pstride;
       for (int cell = slice; cell < pcells; cell += nslices) {
+        // branchless: g == 0 cells contribute 0 to every accumulator, and
+        // their stored argmax still addresses in-bounds LDS — skipping
+        // them would diverge the wave at 16-lane granularity
         const float g = gp[cell];
-        if (g == 0.0f) continue;
         const int sub = (int)sp[cell];
         const int px = cell % PW;
         const int py = cell / PW;
