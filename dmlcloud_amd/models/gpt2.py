@@ -13,6 +13,9 @@ import torch
 from torch import nn
 from torch.nn import functional as F
 
+from ..ops.fused_ln import LayerNorm
+from ..ops.fused_loss import cross_entropy
+
 
 @dataclass
 class GPT2Config:
@@ -47,9 +50,9 @@ class CausalSelfAttention(nn.Module):
 class Block(nn.Module):
     def __init__(self, cfg: GPT2Config):
         super().__init__()
-        self.ln_1 = nn.LayerNorm(cfg.n_embd)
+        self.ln_1 = LayerNorm(cfg.n_embd)
         self.attn = CausalSelfAttention(cfg)
-        self.ln_2 = nn.LayerNorm(cfg.n_embd)
+        self.ln_2 = LayerNorm(cfg.n_embd)
         self.mlp = nn.Sequential(
             nn.Linear(cfg.n_embd, 4 * cfg.n_embd),
             nn.GELU(approximate='tanh'),
@@ -70,7 +73,7 @@ class GPT2(nn.Module):
         self.wte = nn.Embedding(cfg.vocab_size, cfg.n_embd)
         self.wpe = nn.Embedding(cfg.n_positions, cfg.n_embd)
         self.blocks = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layer))
-        self.ln_f = nn.LayerNorm(cfg.n_embd)
+        self.ln_f = LayerNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)
         self.lm_head.weight = self.wte.weight  # tied
 
@@ -99,7 +102,8 @@ class GPT2(nn.Module):
         logits = self.lm_head(x)
         if targets is None:
             return logits
-        loss = F.cross_entropy(logits.view(-1, logits.size(-1)), targets.reshape(-1))
+        # fused online-softmax CE on gfx950 for bf16 logits
+        loss = cross_entropy(logits.view(-1, logits.size(-1)), targets.reshape(-1))
         return logits, loss
 
 

@@ -40,6 +40,18 @@ void conv3x3_relu_pool_bwd_data(at::Tensor dpooled, at::Tensor argmax, at::Tenso
 void conv3x3_relu_pool_bwd_weight(at::Tensor dpooled, at::Tensor argmax, at::Tensor pooled,
                                   at::Tensor in, at::Tensor dw, at::Tensor db);
 
+// layernorm.hip
+void layernorm_fwd(at::Tensor x, at::Tensor gamma, at::Tensor beta, at::Tensor y,
+                   at::Tensor mean, at::Tensor rstd, double eps);
+void layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor mean, at::Tensor rstd,
+                   at::Tensor gamma, at::Tensor dx, at::Tensor dgb_ws, at::Tensor dgamma,
+                   at::Tensor dbeta);
+
+// loss.hip
+void ce_fwd(at::Tensor logits, at::Tensor targets, at::Tensor loss, at::Tensor lse);
+void ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Tensor scale,
+            at::Tensor dlogits);
+
 } // namespace dmlamd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -65,4 +77,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused conv+relu+pool input gradient");
   m.def("conv3x3_relu_pool_bwd_weight", &dmlamd::conv3x3_relu_pool_bwd_weight,
         "Fused conv+relu+pool weight/bias gradient");
+  m.def("layernorm_fwd", &dmlamd::layernorm_fwd, "Fused bf16 LayerNorm forward (row per wave)");
+  m.def("layernorm_bwd", &dmlamd::layernorm_bwd,
+        "Fused bf16 LayerNorm backward (dx + dgamma/dbeta)");
+  m.def("ce_fwd", &dmlamd::ce_fwd, "Online softmax cross-entropy forward (per-row loss + lse)");
+  m.def("ce_bwd", &dmlamd::ce_bwd, "Cross-entropy backward (dlogits in one pass)");
 }

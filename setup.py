@@ -24,6 +24,8 @@ ext = CUDAExtension(
         os.path.join(CSRC, 'copy.hip'),
         os.path.join(CSRC, 'optim.hip'),
         os.path.join(CSRC, 'smallcnn.hip'),
+        os.path.join(CSRC, 'layernorm.hip'),
+        os.path.join(CSRC, 'loss.hip'),
     ],
     extra_compile_args={
         'cxx': ['-O3', '-std=c++17'],

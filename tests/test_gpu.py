@@ -370,6 +370,95 @@ class TestFusedCnn:
         torch.testing.assert_close(fused.layer2.weight.grad, eager[3].weight.grad, rtol=1e-4, atol=1e-5)
 
 
+class TestFusedLayerNorm:
+    @pytest.mark.parametrize('rows,d', [(64, 768), (1000, 768), (128, 1024), (37, 256)])
+    def test_forward_matches_torch(self, rows, d):
+        import torch.nn.functional as F
+
+        torch.manual_seed(0)
+        x = torch.randn(rows, d, device=DEV).to(torch.bfloat16)
+        g = torch.randn(d, device=DEV).to(torch.bfloat16)
+        b = torch.randn(d, device=DEV).to(torch.bfloat16)
+
+        from dmlcloud_amd import _C
+
+        y = torch.empty_like(x)
+        mean = torch.empty(rows, dtype=torch.float32, device=DEV)
+        rstd = torch.empty(rows, dtype=torch.float32, device=DEV)
+        _C.layernorm_fwd(x, g, b, y, mean, rstd, 1e-5)
+        ref_out = F.layer_norm(x.float(), (d,), g.float(), b.float(), 1e-5).to(torch.bfloat16)
+        torch.testing.assert_close(y.float(), ref_out.float(), rtol=2e-2, atol=2e-2)
+
+    def test_backward_matches_torch(self):
+        import torch.nn.functional as F
+
+        from dmlcloud_amd.ops.fused_ln import LayerNorm
+
+        torch.manual_seed(1)
+        rows, d = 512, 768
+        ln = LayerNorm(d).to(DEV).to(torch.bfloat16)
+        x = torch.randn(rows, d, device=DEV).to(torch.bfloat16).requires_grad_(True)
+        out = ln(x)
+        dy = torch.randn_like(out)
+        out.backward(dy)
+
+        x2 = x.detach().float().requires_grad_(True)
+        g2 = ln.weight.detach().float().requires_grad_(True)
+        b2 = ln.bias.detach().float().requires_grad_(True)
+        ref = F.layer_norm(x2, (d,), g2, b2, 1e-5)
+        ref.backward(dy.float())
+
+        torch.testing.assert_close(x.grad.float(), x2.grad, rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(ln.weight.grad.float(), g2.grad, rtol=2e-2, atol=1.0)
+        torch.testing.assert_close(ln.bias.grad.float(), b2.grad, rtol=2e-2, atol=1.0)
+
+
+class TestFusedCE:
+    @pytest.mark.parametrize('rows,v', [(128, 50257), (64, 512), (256, 1000)])
+    def test_loss_matches_torch(self, rows, v):
+        import torch.nn.functional as F
+
+        from dmlcloud_amd.ops.fused_loss import cross_entropy
+
+        torch.manual_seed(0)
+        logits = (torch.randn(rows, v, device=DEV) * 3).to(torch.bfloat16)
+        targets = torch.randint(0, v, (rows,), device=DEV)
+        loss = cross_entropy(logits, targets)
+        ref = F.cross_entropy(logits.float(), targets)
+        assert loss.item() == pytest.approx(ref.item(), rel=2e-3)
+
+    def test_grad_matches_torch(self):
+        import torch.nn.functional as F
+
+        from dmlcloud_amd.ops.fused_loss import cross_entropy
+
+        torch.manual_seed(1)
+        rows, v = 64, 50257
+        logits = (torch.randn(rows, v, device=DEV) * 2).to(torch.bfloat16).requires_grad_(True)
+        targets = torch.randint(0, v, (rows,), device=DEV)
+        cross_entropy(logits, targets).backward()
+
+        l2 = logits.detach().float().requires_grad_(True)
+        F.cross_entropy(l2, targets).backward()
+        torch.testing.assert_close(logits.grad.float(), l2.grad, rtol=5e-2, atol=1e-5)
+
+    def test_gpt2_tiny_step_matches_eager(self):
+        """Full fused-GPT2 (LN + CE) loss matches a plain-torch computation."""
+        from dmlcloud_amd.models import gpt2_tiny
+
+        torch.manual_seed(0)
+        model = gpt2_tiny().to(DEV)
+        # run fp32 (fused paths disabled) vs bf16 (fused) on same weights
+        idx = torch.randint(0, model.cfg.vocab_size, (2, 32), device=DEV)
+        _, loss32 = model(idx, targets=idx)
+
+        model16 = gpt2_tiny().to(DEV)
+        model16.load_state_dict(model.state_dict())
+        model16 = model16.to(torch.bfloat16)
+        _, loss16 = model16(idx, targets=idx)
+        assert loss16.item() == pytest.approx(loss32.item(), rel=5e-2)
+
+
 class TestPipelineGPU:
     def test_smoke_gpu(self, torch_distributed_cuda):
         from dmlcloud_amd import TrainingPipeline, TrainValStage
