@@ -177,33 +177,46 @@ __global__ void __launch_bounds__(kBlock) layernorm_bwd_kernel(
     }
   }
 
-  // one fp32 partial row per wave: dgb_partials[wave_global][2*D]
-  const int64_t wave_global = (int64_t)blockIdx.x * waves_per_block + wave;
-  float* wp = dgb_partials + wave_global * 2 * D;
+  // cross-wave LDS reduction -> ONE fp32 partial row per BLOCK
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = (float*)smem; // [waves_per_block][2*D]
 #pragma unroll
   for (int j = 0; j < kMaxVec; ++j) {
     const int idx = j * kWave + lane;
     if (idx < nvec) {
 #pragma unroll
       for (int k = 0; k < 8; ++k) {
-        wp[idx * 8 + k] = dgamma_acc[j][k];
-        wp[D + idx * 8 + k] = dbeta_acc[j][k];
+        red[wave * 2 * D + idx * 8 + k] = dgamma_acc[j][k];
+        red[wave * 2 * D + D + idx * 8 + k] = dbeta_acc[j][k];
       }
     }
   }
+  __syncthreads();
+  float* bp = dgb_partials + (int64_t)blockIdx.x * 2 * D;
+  for (int i = threadIdx.x; i < 2 * D; i += kBlock) {
+    float total = 0.0f;
+#pragma unroll
+    for (int w = 0; w < kBlock / kWave; ++w) total += red[w * 2 * D + i];
+    bp[i] = total;
+  }
 }
 
-// Reduce the per-wave partials per column and emit bf16 grads.
+// Reduce the per-block partial rows per column: one block per column,
+// threads over partial rows, wave+LDS tree, one bf16 store.
 __global__ void __launch_bounds__(kBlock) dgb_reduce_kernel(
     const float* __restrict__ dgb_partials, __hip_bfloat16* __restrict__ dgamma,
-    __hip_bfloat16* __restrict__ dbeta, int D, int nwaves) {
-  for (int i = threadIdx.x + blockIdx.x * kBlock; i < 2 * D; i += gridDim.x * kBlock) {
-    float total = 0.0f;
-    for (int w = 0; w < nwaves; ++w) total += dgb_partials[(int64_t)w * 2 * D + i];
-    if (i < D) {
-      dgamma[i] = __float2bfloat16(total);
+    __hip_bfloat16* __restrict__ dbeta, int D, int nrows) {
+  const int col = blockIdx.x; // [0, 2*D)
+  float local = 0.0f;
+  for (int r = threadIdx.x; r < nrows; r += kBlock) {
+    local += dgb_partials[(int64_t)r * 2 * D + col];
+  }
+  const float total = block_reduce<float, OP_SUM>(local);
+  if (threadIdx.x == 0) {
+    if (col < D) {
+      dgamma[col] = __float2bfloat16(total);
     } else {
-      dbeta[i - D] = __float2bfloat16(total);
+      dbeta[col - D] = __float2bfloat16(total);
     }
   }
 }
@@ -236,19 +249,20 @@ void layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor mean, at::Tensor rstd
   const int waves_per_block = kBlock / kWave;
   const int blocks =
       (int)std::min<int64_t>((R + waves_per_block - 1) / waves_per_block, kLnBwdBlocks);
-  const int nwaves = blocks * waves_per_block;
-  TORCH_CHECK(dgb_ws.numel() >= (int64_t)nwaves * 2 * D && dgb_ws.scalar_type() == at::kFloat,
+  TORCH_CHECK(dgb_ws.numel() >= (int64_t)blocks * 2 * D && dgb_ws.scalar_type() == at::kFloat,
               "dgb workspace too small");
-  // no memset needed: every launched wave (idle ones included) writes its
-  // full partial row (zeros if it processed no rows)
-  hipLaunchKernelGGL(layernorm_bwd_kernel, dim3(blocks), dim3(kBlock), 0, stream,
+  const int lds_bytes = waves_per_block * 2 * D * (int)sizeof(float);
+  TORCH_CHECK(lds_bytes <= 160 * 1024, "D too large for the cross-wave reduction");
+  // no memset needed: every launched wave (idle ones included) contributes
+  // zeros, and every block writes its full partial row
+  hipLaunchKernelGGL(layernorm_bwd_kernel, dim3(blocks), dim3(kBlock), lds_bytes, stream,
                      (const __hip_bfloat16*)dy.data_ptr(), (const __hip_bfloat16*)x.data_ptr(),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
                      (const __hip_bfloat16*)gamma.data_ptr(), (__hip_bfloat16*)dx.data_ptr(),
                      dgb_ws.data_ptr<float>(), R, D);
-  hipLaunchKernelGGL(dgb_reduce_kernel, dim3(grid_for(2 * D, kBlock)), dim3(kBlock), 0, stream,
+  hipLaunchKernelGGL(dgb_reduce_kernel, dim3(2 * D), dim3(kBlock), 0, stream,
                      dgb_ws.data_ptr<float>(), (__hip_bfloat16*)dgamma.data_ptr(),
-                     (__hip_bfloat16*)dbeta.data_ptr(), D, nwaves);
+                     (__hip_bfloat16*)dbeta.data_ptr(), D, blocks);
 }
 
 } // namespace dmlamd

@@ -31,9 +31,8 @@ class _LayerNormFn(torch.autograd.Function):
         dy = dy.contiguous()
         D = x.shape[-1]
         dx = torch.empty_like(x)
-        # per-wave dgamma/dbeta partials (256 blocks x 4 waves, see
-        # layernorm.hip kLnBwdBlocks); the kernel overwrites every row
-        dgb_ws = torch.empty(1024 * 2 * D, dtype=torch.float32, device=x.device)
+        # per-block dgamma/dbeta partial rows (kLnBwdBlocks in layernorm.hip)
+        dgb_ws = torch.empty(256 * 2 * D, dtype=torch.float32, device=x.device)
         dgamma = torch.empty(D, dtype=x.dtype, device=x.device)
         dbeta = torch.empty(D, dtype=x.dtype, device=x.device)
         _C.layernorm_bwd(dy, x, mean, rstd, weight, dx, dgb_ws, dgamma, dbeta)
