@@ -61,12 +61,10 @@ __host__ __device__ __forceinline__ PlaneGeom plane_geom(int H, int W) {
   return g;
 }
 
-// Zero a padded LDS region then fill its interior with input planes.
-__device__ __forceinline__ void stage_plane_padded(const float* __restrict__ g,
-                                                   float* __restrict__ lds, int CIN, int H,
-                                                   int W, const PlaneGeom& pg) {
-  for (int i = threadIdx.x; i < CIN * pg.cs; i += kBlock) lds[i] = 0.0f;
-  __syncthreads();
+// Fill the interior of an (already zeroed) padded LDS region.
+__device__ __forceinline__ void fill_plane_padded(const float* __restrict__ g,
+                                                  float* __restrict__ lds, int CIN, int H,
+                                                  int W, const PlaneGeom& pg) {
   const int plane = H * W;
   for (int i = threadIdx.x; i < CIN * plane; i += kBlock) {
     const int ci = i / plane;
@@ -112,9 +110,11 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_fwd_kernel(
   const int nblocks = COUT * BH * BW;
 
   stage_to_lds(w, wlds, COUT * CIN * 9);
+  for (int i = threadIdx.x; i < CIN * pg.cs; i += kBlock) ilds[i] = 0.0f; // halo, once
+  __syncthreads();
 
   for (int n = blockIdx.x; n < N; n += gridDim.x) {
-    stage_plane_padded(in + (int64_t)n * plane, ilds, CIN, H, W, pg);
+    fill_plane_padded(in + (int64_t)n * plane, ilds, CIN, H, W, pg);
     __syncthreads();
 
     for (int t = threadIdx.x; t < nblocks; t += kBlock) {
@@ -312,16 +312,27 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_weight_kernel(
   float acc[9] = {0, 0, 0, 0, 0, 0, 0, 0, 0};
   float accb = 0.0f;
 
+  // zero the input halo ONCE; interiors are overwritten every sample
+  for (int i = threadIdx.x; i < CIN * pg.cs; i += kBlock) ilds[i] = 0.0f;
+  __syncthreads();
+
   for (int s = 0; s < nvalid; ++s) {
     const int64_t n = n0 + s;
-    stage_plane_padded(in + n * iplane, ilds, CIN, H, W, pg);
+    fill_plane_padded(in + n * iplane, ilds, CIN, H, W, pg);
     for (int cell = threadIdx.x; cell < cells; cell += kBlock) {
       const int64_t pidx = n * cells + cell;
       const int cco = cell / pcells;
       const int rem = cell - cco * pcells;
       const float pv = pooled[pidx];
       glds[cco * pstride + rem] = (pv > 0.0f) ? dpooled[pidx] : 0.0f;
-      slds[cco * pstride + rem] = (float)argmax[pidx];
+      // pre-decode the argmax into a padded-plane offset so the hot loop
+      // below needs no div/mod per cell (offsets < 2048 are exact floats)
+      const int sub = argmax[pidx];
+      const int px = rem % PW;
+      const int py = rem / PW;
+      const int yy = 2 * py + (sub >> 1);
+      const int xx = 2 * px + (sub & 1);
+      slds[cco * pstride + rem] = (float)(yy * pg.PADW + xx);
     }
     __syncthreads();
 
@@ -331,17 +342,13 @@ __global__ void __launch_bounds__(kBlock) conv3x3_relu_pool_bwd_weight_kernel(
       const float* sp = slds + co * pstride;
       for (int cell = slice; cell < pcells; cell += nslices) {
         // branchless: g == 0 cells contribute 0 to every accumulator, and
-        // their stored argmax still addresses in-bounds LDS — skipping
+        // their stored offset still addresses in-bounds LDS — skipping
         // them would diverge the wave at 16-lane granularity
         const float g = gp[cell];
-        const int sub = (int)sp[cell];
-        const int px = cell % PW;
-        const int py = cell / PW;
-        const int yy = 2 * py + (sub >> 1);
-        const int xx = 2 * px + (sub & 1);
+        const int off = (int)sp[cell];
         if (ci == 0) accb += g;
         // input rows yy-1..yy+1 -> padded rows yy..yy+2, cols likewise
-        const float* iw = ip + yy * pg.PADW + xx;
+        const float* iw = ip + off;
 #pragma unroll
         for (int ky = 0; ky < 3; ++ky) {
 #pragma unroll
