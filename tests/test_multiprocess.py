@@ -149,6 +149,30 @@ def _flat_matches_ddp_math(rank, world_size, tmpdir):
         torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-6)
 
 
+def _flat_bf16_sync(rank, world_size, tmpdir):
+    """bf16 flat replica: broadcast + synced mixed-precision step keeps
+    ranks bitwise identical."""
+    from dmlcloud_amd.parallel import FlatAdam, FlatReplica
+
+    torch.manual_seed(rank)
+    model = torch.nn.Linear(8, 4)
+    replica = FlatReplica(model, dtype=torch.bfloat16)
+    opt = FlatAdam(replica, lr=1e-2)
+
+    x = torch.randn(4, 8, generator=torch.Generator().manual_seed(50 + rank)).to(torch.bfloat16)
+    replica.zero_grad()
+    replica(x).float().pow(2).mean().backward()
+    replica.grad_sync()
+    opt.step()
+
+    gathered = [torch.empty_like(replica.flat_param) for _ in range(world_size)]
+    dist.all_gather(gathered, replica.flat_param)
+    torch.testing.assert_close(gathered[0], gathered[1])
+    gm = [torch.empty_like(replica.flat_master) for _ in range(world_size)]
+    dist.all_gather(gm, replica.flat_master)
+    torch.testing.assert_close(gm[0], gm[1])
+
+
 def _pipeline_two_ranks(rank, world_size, tmpdir):
     from dmlcloud_amd import TrainingPipeline, TrainValStage
 
@@ -211,6 +235,7 @@ def _root_helpers(rank, world_size, tmpdir):
         '_tracker_fused',
         '_flat_replica_sync',
         '_flat_matches_ddp_math',
+        '_flat_bf16_sync',
         '_pipeline_two_ranks',
         '_root_helpers',
     ],
