@@ -94,7 +94,12 @@ class FlatReplica:
         self.flat_param, self.flat_grad, self.flat_master = _flatten_params(self.params, dtype=dtype)
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         if broadcast and dist.is_initialized() and self.world_size > 1:
-            dist.broadcast(self.flat_param, src=0, group=process_group)
+            if self.flat_master is not None:
+                # the fp32 master is authoritative; params re-derive from it
+                dist.broadcast(self.flat_master, src=0, group=process_group)
+                self.flat_param.copy_(self.flat_master.to(dtype))
+            else:
+                dist.broadcast(self.flat_param, src=0, group=process_group)
             # buffers (e.g. BN running stats) follow rank 0 once at init
             for buf in module.buffers():
                 dist.broadcast(buf, src=0, group=process_group)
