@@ -56,6 +56,7 @@ def parse_args():
     p.add_argument('--batch-size', type=int, default=None, help='per-GPU batch size')
     p.add_argument('--no-graph', action='store_true', help='disable hipGraph capture')
     p.add_argument('--no-fused', action='store_true', help='disable the fused smallcnn kernels (MIOpen path)')
+    p.add_argument('--tunableop', action='store_true', help='enable PyTorch TunableOp GEMM tuning (hipBLASLt)')
     p.add_argument('--no-channels-last', action='store_true', help='disable NHWC memory format (resnet50)')
     p.add_argument('--seq-len', type=int, default=1024, help='gpt2 sequence length')
     return p.parse_args()
@@ -191,6 +192,13 @@ class BenchStage(TrainValStage):
 
 def main():
     args = parse_args()
+
+    if args.tunableop:
+        # hipBLASLt/rocBLAS algorithm search for this config's GEMM shapes;
+        # tuning happens during warmup, results cached per rank
+        os.environ.setdefault('PYTORCH_TUNABLEOP_ENABLED', '1')
+        os.environ.setdefault('PYTORCH_TUNABLEOP_TUNING', '1')
+        os.environ.setdefault('PYTORCH_TUNABLEOP_FILENAME', f'gpurun_out/tunableop_{os.environ.get("RANK", 0)}.csv')
 
     if 'MASTER_ADDR' not in os.environ and args.gpus == 1:
         pass  # dummy single-process group below

@@ -22,6 +22,9 @@ from typing import Any, Optional
 
 
 def _format_cell(value: Any, width: int) -> str:
+    # unwrap 0-dim / 1-element tensors (duck-typed; no torch import here)
+    if hasattr(value, 'numel') and hasattr(value, 'item') and value.numel() == 1:
+        value = value.item()
     if value is None:
         text = ''
     elif isinstance(value, float):
