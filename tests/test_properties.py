@@ -1,0 +1,81 @@
+"""Property-based tests (hypothesis) for the sharding math and config."""
+
+import sys
+
+import pytest
+
+try:
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    HAS_HYPOTHESIS = True
+except ImportError:
+    HAS_HYPOTHESIS = False
+
+if not HAS_HYPOTHESIS:  # pragma: no cover
+    pytest.skip('hypothesis not installed', allow_module_level=True)
+
+from dmlcloud_amd.config import Config
+from dmlcloud_amd.data import chunk_and_shard_indices, shard_indices
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    n=st.integers(0, 500),
+    world=st.integers(1, 16),
+    shuffle=st.booleans(),
+    seed=st.integers(0, 10),
+)
+def test_shard_indices_partition(n, world, shuffle, seed):
+    """With even_shards, shards partition the first n - n%world elements
+    into equal disjoint sets."""
+    shards = [shard_indices(n, r, world, shuffle=shuffle, even_shards=True, seed=seed) for r in range(world)]
+    sizes = {len(s) for s in shards}
+    assert len(sizes) == 1  # equal shards
+    combined = sorted(i for s in shards for i in s)
+    kept = n - n % world
+    assert combined == sorted(range(n))[:kept] if not shuffle else len(combined) == kept
+    assert len(set(combined)) == len(combined)  # disjoint
+
+
+@settings(max_examples=200, deadline=None)
+@given(n=st.integers(0, 500), world=st.integers(1, 16))
+def test_shard_indices_uneven_cover(n, world):
+    """Without even_shards, shards cover ALL elements disjointly."""
+    shards = [shard_indices(n, r, world, even_shards=False) for r in range(world)]
+    combined = sorted(i for s in shards for i in s)
+    assert combined == list(range(n))
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    n=st.integers(1, 1000),
+    chunk=st.integers(1, 50),
+    world=st.integers(1, 8),
+    overlap=st.integers(0, 10),
+)
+def test_chunks_within_bounds(n, chunk, world, overlap):
+    """equal_chunks windows start within the data and have uniform length."""
+    for r in range(world):
+        for start, end in chunk_and_shard_indices(n, chunk, r, world, chunk_overlap=overlap, equal_chunks=True):
+            assert 0 <= start < n or n < chunk  # starts in-range when any chunk exists
+            assert end - start == chunk + overlap
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    data=st.dictionaries(
+        st.text(st.characters(whitelist_categories=('Ll',), whitelist_characters='_'), min_size=1, max_size=8),
+        st.one_of(st.integers(), st.floats(allow_nan=False), st.text(max_size=10), st.booleans()),
+        max_size=6,
+    )
+)
+def test_config_yaml_roundtrip(data, tmp_path_factory):
+    cfg = Config.create(data)
+    path = tmp_path_factory.mktemp('cfg') / 'c.yaml'
+    cfg.save(path)
+    assert Config.load(path) == cfg
+
+
+if __name__ == '__main__':
+    sys.exit(pytest.main([__file__]))
