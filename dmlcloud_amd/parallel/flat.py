@@ -56,7 +56,7 @@ def _flatten_params(params: List[torch.nn.Parameter], dtype=torch.float32, devic
         flat[o : o + n].copy_(fp32.to(dtype))
         p.data = flat[o : o + n].view(p.shape)
         p.grad = grad_flat[o : o + n].view(p.shape)
-    return flat, grad_flat, master
+    return flat, grad_flat, master, offsets
 
 
 class FlatReplica:
@@ -77,11 +77,19 @@ class FlatReplica:
         process_group=None,
         broadcast: bool = True,
         dtype: torch.dtype = torch.float32,
+        overlap_buckets_mb: Optional[int] = None,
     ):
         """dtype=torch.bfloat16 enables true mixed precision: the model
         computes in bf16 (no per-layer autocast casts), gradients
         all-reduce in bf16 (half the xGMI bytes), and the fused optimizer
-        holds the fp32 master (replica.flat_master)."""
+        holds the fp32 master (replica.flat_master).
+
+        overlap_buckets_mb: when set, gradients all-reduce bucket-by-bucket
+        DURING backward (post-accumulate-grad hooks fire async collectives
+        in backward-completion order over contiguous flat segments) and
+        grad_sync() just waits — overlapping communication with the rest
+        of backward like DDP does, but over the flat buffer. Leave None
+        for the single-collective mode (hipGraph-capturable)."""
         if dtype not in (torch.float32, torch.bfloat16):
             raise ValueError('FlatReplica supports fp32 or bf16 parameter buffers')
         self.module = module
@@ -91,8 +99,15 @@ class FlatReplica:
         for p in self.params:
             if p.dtype != torch.float32:
                 raise ValueError('FlatReplica expects fp32 module parameters at construction')
-        self.flat_param, self.flat_grad, self.flat_master = _flatten_params(self.params, dtype=dtype)
+        self.flat_param, self.flat_grad, self.flat_master, self._offsets = _flatten_params(
+            self.params, dtype=dtype
+        )
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
+
+        self._buckets = None
+        self._pending_works = []
+        if overlap_buckets_mb:
+            self._setup_overlap(overlap_buckets_mb)
         if broadcast and dist.is_initialized() and self.world_size > 1:
             if self.flat_master is not None:
                 # the fp32 master is authoritative; params re-derive from it
@@ -107,15 +122,73 @@ class FlatReplica:
     def __call__(self, *args, **kwargs):
         return self.module(*args, **kwargs)
 
+    def _setup_overlap(self, bucket_mb: int):
+        """Partition the flat buffer into contiguous buckets ordered by
+        backward completion (params in reverse registration order) and
+        hook each param's gradient accumulation."""
+        bucket_bytes = bucket_mb << 20
+        elsize = self.flat_grad.element_size()
+        buckets = []  # list of dicts: {start, end, params:set, done:set}
+        current = {'params': [], 'bytes': 0}
+        # reverse order: the last registered params finish backward first
+        for i in reversed(range(len(self.params))):
+            current['params'].append(i)
+            current['bytes'] += self.params[i].numel() * elsize
+            if current['bytes'] >= bucket_bytes:
+                buckets.append(current['params'])
+                current = {'params': [], 'bytes': 0}
+        if current['params']:
+            buckets.append(current['params'])
+
+        self._buckets = []
+        self._param_bucket = {}
+        for param_ids in buckets:
+            start = min(self._offsets[i] for i in param_ids)
+            end = max(
+                self._offsets[i] + self.params[i].numel() for i in param_ids
+            )
+            bucket = {'segment': self.flat_grad[start:end], 'params': set(param_ids), 'done': set()}
+            self._buckets.append(bucket)
+            for i in param_ids:
+                self._param_bucket[i] = bucket
+
+        for i, p in enumerate(self.params):
+            p.register_post_accumulate_grad_hook(self._make_hook(i))
+
+    def _make_hook(self, index: int):
+        def hook(_param):
+            if self.world_size <= 1:
+                return
+            bucket = self._param_bucket[index]
+            bucket['done'].add(index)
+            if bucket['done'] == bucket['params']:
+                bucket['done'] = set()
+                work = dist.all_reduce(bucket['segment'], group=self.process_group, async_op=True)
+                self._pending_works.append(work)
+
+        return hook
+
     def zero_grad(self, set_to_none: bool = False):
         # one memset kernel; set_to_none is meaningless for a flat buffer
         self.flat_grad.zero_()
 
     def grad_sync(self):
-        """All-reduce the flat gradient buffer (SUM; averaging happens in
-        the optimizer's grad_scale)."""
-        if self.world_size > 1:
+        """Make gradients globally consistent. Single-collective mode:
+        ONE all-reduce of the flat buffer. Overlap mode: wait on the
+        bucket collectives that backward already launched."""
+        if self.world_size <= 1:
+            return
+        if self._buckets is None:
             dist.all_reduce(self.flat_grad, group=self.process_group)
+            return
+        for work in self._pending_works:
+            work.wait()
+        self._pending_works.clear()
+        for bucket in self._buckets:
+            # a bucket whose params got no grads this step never fired
+            if bucket['done']:
+                bucket['done'] = set()
+                dist.all_reduce(bucket['segment'], group=self.process_group)
 
     @property
     def grad_scale(self) -> float:

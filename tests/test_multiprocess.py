@@ -173,6 +173,32 @@ def _flat_bf16_sync(rank, world_size, tmpdir):
     torch.testing.assert_close(gm[0], gm[1])
 
 
+def _flat_overlap_matches_single(rank, world_size, tmpdir):
+    """Bucketed-overlap grad sync == single-collective grad sync."""
+    from dmlcloud_amd.parallel import FlatReplica, FlatSGD
+
+    def build(overlap):
+        torch.manual_seed(0)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(16, 64), torch.nn.ReLU(), torch.nn.Linear(64, 64), torch.nn.Linear(64, 4)
+        )
+        replica = FlatReplica(model, overlap_buckets_mb=overlap)
+        return replica, FlatSGD(replica, lr=0.1)
+
+    rep_a, opt_a = build(None)
+    rep_b, opt_b = build(1)  # tiny buckets -> several collectives
+    assert rep_b._buckets is not None and len(rep_b._buckets) >= 1
+
+    for it in range(3):
+        x = torch.randn(8, 16, generator=torch.Generator().manual_seed(it * 10 + rank))
+        for rep, opt in ((rep_a, opt_a), (rep_b, opt_b)):
+            rep.zero_grad()
+            rep(x).pow(2).mean().backward()
+            rep.grad_sync()
+            opt.step()
+    torch.testing.assert_close(rep_a.flat_param, rep_b.flat_param, rtol=1e-6, atol=1e-7)
+
+
 def _pipeline_two_ranks(rank, world_size, tmpdir):
     from dmlcloud_amd import TrainingPipeline, TrainValStage
 
@@ -236,6 +262,7 @@ def _root_helpers(rank, world_size, tmpdir):
         '_flat_replica_sync',
         '_flat_matches_ddp_math',
         '_flat_bf16_sync',
+        '_flat_overlap_matches_single',
         '_pipeline_two_ranks',
         '_root_helpers',
     ],
