@@ -1,0 +1,56 @@
+"""bench.py driver contract: runs standalone, emits one valid JSON line."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_bench(extra):
+    proc = subprocess.run(
+        [sys.executable, 'bench.py', '--steps', '2', '--warmup', '1', '--batch-size', '64'] + extra,
+        cwd=REPO,
+        capture_output=True,
+        text=True,
+        timeout=420,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    lines = [ln for ln in proc.stdout.splitlines() if ln.startswith('{')]
+    assert len(lines) == 1, proc.stdout
+    return json.loads(lines[0])
+
+
+@pytest.mark.parametrize('model', ['mnist'])
+def test_bench_json_contract(model):
+    result = _run_bench(['--model', model])
+    for key in (
+        'metric',
+        'value',
+        'unit',
+        'n_gpus',
+        'steps',
+        'warmup',
+        'ms_per_step',
+        'higher_is_better',
+        'scaling',
+        'vs_baseline',
+        'dtype',
+        'data',
+        'config',
+    ):
+        assert key in result, key
+    assert result['n_gpus'] == 1
+    assert result['steps'] == 2
+    assert result['data'] == 'synthetic'
+    assert result['value'] > 0
+    assert result['config']['parallelism'] == 'dp1'
+    assert result['scaling'] == 'weak'
+
+
+def test_bench_ddp_impl():
+    result = _run_bench(['--impl', 'ddp'])
+    assert result['config']['impl'] == 'ddp'
