@@ -440,8 +440,13 @@ void conv3x3_relu_pool_bwd_weight(at::Tensor dpooled, at::Tensor argmax, at::Ten
       (CIN * pg.cs + 2 * COUT * pstride + kBlock) * (int)sizeof(float);
   TORCH_CHECK(lds_bytes <= kMaxLds, "bwd_weight staging exceeds LDS (", lds_bytes, " B)");
   // chunk so that the grid stays ~>= 512 blocks while cutting atomics
+  // (DMLCLOUD_SMALLCNN_BWDW_CHUNK overrides for tuning experiments)
   int samples = 1;
-  while (samples < 16 && (N + samples * 2 - 1) / (samples * 2) >= 512) samples *= 2;
+  if (const char* env = std::getenv("DMLCLOUD_SMALLCNN_BWDW_CHUNK")) {
+    samples = std::max(1, atoi(env));
+  } else {
+    while (samples < 16 && (N + samples * 2 - 1) / (samples * 2) >= 512) samples *= 2;
+  }
   const int blocks = (N + samples - 1) / samples;
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(conv3x3_relu_pool_bwd_weight_kernel, dim3(blocks), dim3(kBlock), lds_bytes,
