@@ -1,0 +1,183 @@
+// EXPERIMENTAL: flash-attention forward for gfx950 (MFMA bf16, D=64).
+//
+// Correctness-first v1 of the custom attention path (docs/ROADMAP.md
+// item 1; the backward follows in round 2 — this kernel already emits
+// the per-row lse the backward needs). Tiling mirrors
+// ops/_attention_ref.py exactly; fragment maps are the empirically
+// confirmed ones (profiles/mfma_16x16x32_bf16_layout.txt):
+//
+//   A (bf16x8): lane l elem j <-> A[l%16][8*(l/16)+j]
+//   B (bf16x8): lane l elem j <-> B[8*(l/16)+j][l%16]
+//   C/D (f32x4): lane l reg r <-> D[4*(l/16)+r][l%16]
+//
+// Structure: ONE WAVE owns a 16-row Q tile and iterates 32-key KV tiles.
+// S is computed TRANSPOSED (S^T = K Q^T) so each lane's 8 scores share a
+// single query (softmax row reductions = 2 xor-shuffles); P^T goes
+// through a small per-wave LDS slice to re-enter the PV MFMA as the A
+// operand. Online softmax keeps m/sumexp in registers; the O accumulator
+// (16x64) lives in 4 C fragments.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_bf16.h>
+
+#include "ops_common.h"
+
+namespace dmlamd {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr int kAttnD = 64; // head dim (v1: fixed)
+constexpr int kQT = 16; // query rows per wave
+constexpr int kKT = 32; // keys per kv tile
+constexpr int kPStride = 40; // P_lds row stride in bf16 (16B-aligned rows)
+constexpr int kWavesPerBlock = 4;
+
+__global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
+    float* __restrict__ lse, int BH, int N, float scale, bool causal) {
+  __shared__ __hip_bfloat16 p_lds_all[kWavesPerBlock][kQT][kPStride];
+
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wave = threadIdx.x / kWave;
+  const int row16 = lane & 15; // q index inside the tile (and key row for K frags)
+  const int grp = lane >> 4; // 16-lane group 0..3
+  __hip_bfloat16(*p_lds)[kPStride] = p_lds_all[wave];
+
+  const int tiles_per_bh = N / kQT;
+  const int64_t total_tiles = (int64_t)BH * tiles_per_bh;
+
+  for (int64_t tile = (int64_t)blockIdx.x * kWavesPerBlock + wave; tile < total_tiles;
+       tile += (int64_t)gridDim.x * kWavesPerBlock) {
+    const int bh = tile / tiles_per_bh;
+    const int i0 = (tile - (int64_t)bh * tiles_per_bh) * kQT;
+    const __hip_bfloat16* qb = q + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* kb = k + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* vb = v + (int64_t)bh * N * kAttnD;
+
+    // Q fragments: two k-chunks of 32 over D=64
+    bf16x8 qf[2];
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      qf[c] = *(const bf16x8*)(qb + (int64_t)(i0 + row16) * kAttnD + 32 * c + 8 * grp);
+    }
+
+    float m_run = -1e30f; // running max for THIS lane's q (= row16)
+    float s_run = 0.0f; // running sumexp
+    f32x4 o_acc[4]; // O[16 x 64]: d-block db -> cols 16*db + row16
+#pragma unroll
+    for (int db = 0; db < 4; ++db) o_acc[db] = f32x4{0, 0, 0, 0};
+
+    const int kv_end = causal ? (i0 + kQT) : N;
+    for (int j0 = 0; j0 < kv_end; j0 += kKT) {
+      // ---- S^T = K Q^T for two 16-key halves ----
+      float sv[8]; // h*4 + r: key = j0 + 16h + 4grp + r, query = row16
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        f32x4 acc = {0, 0, 0, 0};
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          const bf16x8 kf =
+              *(const bf16x8*)(kb + (int64_t)(j0 + 16 * h + row16) * kAttnD + 32 * c + 8 * grp);
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[c], acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float s = acc[r] * scale;
+          if (causal) {
+            const int key_g = j0 + 16 * h + 4 * grp + r;
+            const int q_g = i0 + row16;
+            if (key_g > q_g) s = -1e30f;
+          }
+          sv[h * 4 + r] = s;
+        }
+      }
+
+      // ---- online softmax (per q = row16; reduce across grp groups) ----
+      float mt = sv[0];
+#pragma unroll
+      for (int x = 1; x < 8; ++x) mt = fmaxf(mt, sv[x]);
+      mt = fmaxf(mt, __shfl_xor(mt, 16, kWave));
+      mt = fmaxf(mt, __shfl_xor(mt, 32, kWave));
+      const float m_new = fmaxf(m_run, mt);
+      const float alpha = __expf(m_run - m_new); // m_run=-1e30 -> exp(-inf)=0
+
+      float ps = 0.0f;
+#pragma unroll
+      for (int x = 0; x < 8; ++x) {
+        sv[x] = __expf(sv[x] - m_new);
+        ps += sv[x];
+      }
+      ps += __shfl_xor(ps, 16, kWave);
+      ps += __shfl_xor(ps, 32, kWave);
+      s_run = s_run * alpha + ps;
+      m_run = m_new;
+
+      // ---- P^T -> LDS (per-wave slice; same-wave visibility only) ----
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          p_lds[row16][16 * h + 4 * grp + r] = __float2bfloat16(sv[h * 4 + r]);
+        }
+      }
+      // same-wave LDS read-after-write: compiler inserts lgkmcnt waits
+
+      // ---- PV: A = P[q][key] from LDS, B = V[key][d] ----
+      const bf16x8 pf = *(const bf16x8*)(&p_lds[row16][8 * grp]);
+#pragma unroll
+      for (int db = 0; db < 4; ++db) {
+        bf16x8 vf;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          vf[j] = *(const __bf16*)(vb + (int64_t)(j0 + 8 * grp + j) * kAttnD + 16 * db + row16);
+        }
+        // rescale THIS accumulator's rows by alpha (row q_o = 4*grp + r)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float a_o = __shfl(alpha, 4 * grp + r, kWave); // lane q_o holds alpha[q_o]
+          o_acc[db][r] *= a_o;
+        }
+        o_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, o_acc[db], 0, 0, 0);
+      }
+    }
+
+    // ---- epilogue: O /= sumexp, store; lse = m + log(sumexp) ----
+#pragma unroll
+    for (int db = 0; db < 4; ++db) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int q_o = 4 * grp + r;
+        const float denom = __shfl(s_run, q_o, kWave);
+        const float val = o_acc[db][r] / denom;
+        o[(int64_t)bh * N * kAttnD + (int64_t)(i0 + q_o) * kAttnD + 16 * db + row16] =
+            __float2bfloat16(val);
+      }
+    }
+    if (lane < 16) {
+      lse[(int64_t)bh * N + i0 + row16] = m_run + __logf(s_run);
+    }
+  }
+}
+
+void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor lse,
+              double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 && q.is_contiguous(),
+              "q must be contiguous bf16 [B,H,N,D]");
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == kAttnD, "v1 supports head dim 64");
+  const int B = q.size(0), H = q.size(1), N = q.size(2);
+  TORCH_CHECK(N % kKT == 0, "N must be a multiple of 32");
+  const int BH = B * H;
+  const int64_t total_tiles = (int64_t)BH * (N / kQT);
+  const int blocks =
+      (int)std::min<int64_t>((total_tiles + kWavesPerBlock - 1) / kWavesPerBlock, kMaxGrid);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(attn_fwd_kernel, dim3(blocks), dim3(kWavesPerBlock * kWave), 0, stream,
+                     (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                     (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)o.data_ptr(),
+                     lse.data_ptr<float>(), BH, N, (float)scale, causal);
+}
+
+} // namespace dmlamd

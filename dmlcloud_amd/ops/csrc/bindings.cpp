@@ -52,6 +52,10 @@ void ce_fwd(at::Tensor logits, at::Tensor targets, at::Tensor loss, at::Tensor l
 void ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Tensor scale,
             at::Tensor dlogits);
 
+// attention.hip (experimental)
+void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor lse,
+              double scale, bool causal);
+
 } // namespace dmlamd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -82,4 +86,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused bf16 LayerNorm backward (dx + dgamma/dbeta)");
   m.def("ce_fwd", &dmlamd::ce_fwd, "Online softmax cross-entropy forward (per-row loss + lse)");
   m.def("ce_bwd", &dmlamd::ce_bwd, "Cross-entropy backward (dlogits in one pass)");
+  m.def("attn_fwd", &dmlamd::attn_fwd,
+        "EXPERIMENTAL flash-attention forward (MFMA bf16, D=64, emits lse)");
 }

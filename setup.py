@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, 'smallcnn.hip'),
         os.path.join(CSRC, 'layernorm.hip'),
         os.path.join(CSRC, 'loss.hip'),
+        os.path.join(CSRC, 'attention.hip'),
     ],
     extra_compile_args={
         'cxx': ['-O3', '-std=c++17'],

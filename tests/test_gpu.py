@@ -459,6 +459,51 @@ class TestFusedCE:
         assert loss16.item() == pytest.approx(loss32.item(), rel=5e-2)
 
 
+class TestAttnFwdExperimental:
+    """EXPERIMENTAL flash-attention forward (attention.hip) vs SDPA and
+    the CPU tile blueprint."""
+
+    @pytest.mark.parametrize('causal', [True, False])
+    @pytest.mark.parametrize('b,h,n', [(1, 1, 64), (2, 3, 128), (2, 4, 1024)])
+    def test_matches_sdpa(self, causal, b, h, n):
+        import math
+
+        from dmlcloud_amd import _C
+
+        torch.manual_seed(0)
+        d = 64
+        q = (torch.randn(b, h, n, d, device=DEV) * 0.5).to(torch.bfloat16)
+        k = (torch.randn(b, h, n, d, device=DEV) * 0.5).to(torch.bfloat16)
+        v = (torch.randn(b, h, n, d, device=DEV) * 0.5).to(torch.bfloat16)
+        o = torch.empty_like(q)
+        lse = torch.empty(b, h, n, dtype=torch.float32, device=DEV)
+        _C.attn_fwd(q, k, v, o, lse, 1.0 / math.sqrt(d), causal)
+
+        ref = torch.nn.functional.scaled_dot_product_attention(
+            q.float(), k.float(), v.float(), is_causal=causal
+        )
+        torch.testing.assert_close(o.float(), ref, rtol=3e-2, atol=3e-2)
+
+    def test_lse_matches_blueprint(self):
+        import math
+
+        from dmlcloud_amd import _C
+        from dmlcloud_amd.ops._attention_ref import flash_attn_fwd_tiled
+
+        torch.manual_seed(1)
+        n, d = 128, 64
+        q = (torch.randn(1, 1, n, d) * 0.5).to(torch.bfloat16)
+        k = (torch.randn(1, 1, n, d) * 0.5).to(torch.bfloat16)
+        v = (torch.randn(1, 1, n, d) * 0.5).to(torch.bfloat16)
+        o = torch.empty_like(q).to(DEV)
+        lse = torch.empty(1, 1, n, dtype=torch.float32, device=DEV)
+        _C.attn_fwd(q.to(DEV), k.to(DEV), v.to(DEV), o, lse, 1.0 / math.sqrt(d), True)
+
+        ref_o, ref_lse = flash_attn_fwd_tiled(q[0, 0].float(), k[0, 0].float(), v[0, 0].float(), causal=True)
+        torch.testing.assert_close(lse.cpu()[0, 0], ref_lse, rtol=1e-2, atol=1e-2)
+        torch.testing.assert_close(o.float().cpu()[0, 0], ref_o, rtol=3e-2, atol=3e-2)
+
+
 class TestPipelineGPU:
     def test_smoke_gpu(self, torch_distributed_cuda):
         from dmlcloud_amd import TrainingPipeline, TrainValStage
