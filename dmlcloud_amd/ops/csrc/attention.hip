@@ -142,13 +142,20 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
         m_run = m_new;
 
         // ---- P^T -> per-wave LDS slice (same-wave visibility only) ----
+        // 4 consecutive keys pack into one 8-byte write per half
+        typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
 #pragma unroll
         for (int h = 0; h < 2; ++h) {
+          bf16x4 pw;
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            p_lds[row16][16 * h + 4 * grp + r] = __float2bfloat16(sv[h * 4 + r]);
-          }
+          for (int r = 0; r < 4; ++r) pw[r] = (__bf16)sv[h * 4 + r];
+          *(bf16x4*)(&p_lds[row16][16 * h + 4 * grp]) = pw;
         }
+
+        // rescale rows (q_o = 4*grp + r) once; lane q_o holds alpha[q_o]
+        float a_o[4];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) a_o[r] = __shfl(alpha, 4 * grp + r, kWave);
 
         // ---- PV: A = P[q][key] from LDS, B = V[key][d] from LDS ----
         const bf16x8 pf = *(const bf16x8*)(&p_lds[row16][8 * grp]);
@@ -160,10 +167,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
             vf[j] = *(const __bf16*)(&v_lds[8 * grp + j][16 * db + row16]);
           }
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const float a_o = __shfl(alpha, 4 * grp + r, kWave); // lane q_o holds alpha[q_o]
-            o_acc[db][r] *= a_o;
-          }
+          for (int r = 0; r < 4; ++r) o_acc[db][r] *= a_o[r];
           o_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, o_acc[db], 0, 0, 0);
         }
       }
