@@ -47,8 +47,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
     float* __restrict__ lse, int BH, int N, float scale, bool causal) {
   __shared__ __hip_bfloat16 p_lds_all[kWavesPerBlock][kQT][kPStride];
-  __shared__ __hip_bfloat16 k_lds[kKT][kKVStride];
-  __shared__ __hip_bfloat16 v_lds[kKT][kKVStride];
+  __shared__ __hip_bfloat16 k_lds[2][kKT][kKVStride];
+  __shared__ __hip_bfloat16 v_lds[2][kKT][kKVStride];
 
   const int lane = threadIdx.x & (kWave - 1);
   const int wave = threadIdx.x / kWave;
@@ -90,13 +90,25 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     const int kv_end_block = causal ? min(qb0 + qrows_per_block, N) : N;
     const int my_kv_end = causal ? (i0 + kQT) : N;
 
-    for (int j0 = 0; j0 < kv_end_block; j0 += kKT) {
-      // ---- cooperative staging: one bf16x8 per thread per tensor ----
-      *(bf16x8*)(&k_lds[st_row][st_col]) =
-          *(const bf16x8*)(kp + (int64_t)(j0 + st_row) * kAttnD + st_col);
-      *(bf16x8*)(&v_lds[st_row][st_col]) =
-          *(const bf16x8*)(vp + (int64_t)(j0 + st_row) * kAttnD + st_col);
-      __syncthreads();
+    // write-late double buffer (guide §6 G15): the NEXT tile's global
+    // loads stay in flight through the current tile's compute; their
+    // ds_write targets the other buffer just before the single barrier.
+    *(bf16x8*)(&k_lds[0][st_row][st_col]) =
+        *(const bf16x8*)(kp + (int64_t)st_row * kAttnD + st_col);
+    *(bf16x8*)(&v_lds[0][st_row][st_col]) =
+        *(const bf16x8*)(vp + (int64_t)st_row * kAttnD + st_col);
+    __syncthreads();
+
+    const int ntiles = (kv_end_block + kKT - 1) / kKT;
+    for (int jt = 0; jt < ntiles; ++jt) {
+      const int j0 = jt * kKT;
+      const int buf = jt & 1;
+      bf16x8 knext, vnext;
+      const bool has_next = jt + 1 < ntiles;
+      if (has_next) {
+        knext = *(const bf16x8*)(kp + (int64_t)(j0 + kKT + st_row) * kAttnD + st_col);
+        vnext = *(const bf16x8*)(vp + (int64_t)(j0 + kKT + st_row) * kAttnD + st_col);
+      }
 
       if (valid && j0 < my_kv_end) {
         // ---- S^T = K Q^T for two 16-key halves ----
@@ -106,7 +118,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
           f32x4 acc = {0, 0, 0, 0};
 #pragma unroll
           for (int c = 0; c < 2; ++c) {
-            const bf16x8 kf = *(const bf16x8*)(&k_lds[16 * h + row16][32 * c + 8 * grp]);
+            const bf16x8 kf = *(const bf16x8*)(&k_lds[buf][16 * h + row16][32 * c + 8 * grp]);
             acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[c], acc, 0, 0, 0);
           }
 #pragma unroll
@@ -164,14 +176,18 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
           bf16x8 vf;
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            vf[j] = *(const __bf16*)(&v_lds[8 * grp + j][16 * db + row16]);
+            vf[j] = *(const __bf16*)(&v_lds[buf][8 * grp + j][16 * db + row16]);
           }
 #pragma unroll
           for (int r = 0; r < 4; ++r) o_acc[db][r] *= a_o[r];
           o_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, o_acc[db], 0, 0, 0);
         }
       }
-      __syncthreads(); // before the next tile overwrites k_lds/v_lds
+      if (has_next) {
+        *(bf16x8*)(&k_lds[buf ^ 1][st_row][st_col]) = knext;
+        *(bf16x8*)(&v_lds[buf ^ 1][st_row][st_col]) = vnext;
+      }
+      __syncthreads(); // readers of buf done AND buf^1 writes visible
     }
 
     if (valid) {
