@@ -38,10 +38,13 @@ constexpr int kWavesPerBlock = 4;
 // column offset span 16 distinct banks (rows at stride 88 bf16 = 44 dwords)
 constexpr int kKVStride = 88;
 
-// v2: the block's 4 waves own 4 consecutive q-tiles of ONE (b,h) and share
+// v2: the block's 4 waves own consecutive q-tiles of ONE (b,h) and share
 // cooperatively staged K/V LDS tiles (one bf16x8 global load per thread per
 // tile — coalesced — instead of per-lane scattered loads; v1 measured
 // 97-186 TF vs AOTriton's 213-458 at the GPT-2 shape).
+// v5: each wave owns TWO 16-row q sub-tiles (32 q rows): the shared K
+// fragments and the scalar V fragment reads amortize over twice the
+// MFMAs, roughly doubling the per-tile MFMA:overhead ratio.
 __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
@@ -60,35 +63,41 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 8;
 
-  const int qrows_per_block = kWavesPerBlock * kQT; // 64
+  const int qrows_per_block = kWavesPerBlock * 2 * kQT; // 128 (2 sub-tiles per wave)
   const int nqb = (N + qrows_per_block - 1) / qrows_per_block;
   const int64_t total_blocks = (int64_t)BH * nqb;
 
   for (int64_t blk = blockIdx.x; blk < total_blocks; blk += gridDim.x) {
     const int bh = blk / nqb;
     const int qb0 = (blk - (int64_t)bh * nqb) * qrows_per_block;
-    const int i0 = qb0 + wave * kQT;
+    const int i0 = qb0 + wave * 2 * kQT; // this wave's 32 q rows
     const bool valid = i0 < N;
     const __hip_bfloat16* qp = q + (int64_t)bh * N * kAttnD;
     const __hip_bfloat16* kp = k + (int64_t)bh * N * kAttnD;
     const __hip_bfloat16* vp = v + (int64_t)bh * N * kAttnD;
 
-    bf16x8 qf[2];
+    bf16x8 qf[2][2]; // [sub-tile][k-chunk]
     if (valid) {
 #pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        qf[c] = *(const bf16x8*)(qp + (int64_t)(i0 + row16) * kAttnD + 32 * c + 8 * grp);
+      for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+          qf[sub][c] = *(const bf16x8*)(qp + (int64_t)(i0 + 16 * sub + row16) * kAttnD +
+                                        32 * c + 8 * grp);
+        }
       }
     }
 
-    float m_run = -1e30f;
-    float s_run = 0.0f;
-    f32x4 o_acc[4];
+    float m_run[2] = {-1e30f, -1e30f};
+    float s_run[2] = {0.0f, 0.0f};
+    f32x4 o_acc[2][4];
 #pragma unroll
-    for (int db = 0; db < 4; ++db) o_acc[db] = f32x4{0, 0, 0, 0};
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+      for (int db = 0; db < 4; ++db) o_acc[sub][db] = f32x4{0, 0, 0, 0};
 
     const int kv_end_block = causal ? min(qb0 + qrows_per_block, N) : N;
-    const int my_kv_end = causal ? (i0 + kQT) : N;
+    const int my_kv_end = causal ? (i0 + 2 * kQT) : N;
 
     // write-late double buffer (guide §6 G15): the NEXT tile's global
     // loads stay in flight through the current tile's compute; their
@@ -111,76 +120,86 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
       }
 
       if (valid && j0 < my_kv_end) {
-        // ---- S^T = K Q^T for two 16-key halves ----
-        float sv[8]; // h*4 + r: key = j0 + 16h + 4grp + r, query = row16
-#pragma unroll
-        for (int h = 0; h < 2; ++h) {
-          f32x4 acc = {0, 0, 0, 0};
-#pragma unroll
-          for (int c = 0; c < 2; ++c) {
-            const bf16x8 kf = *(const bf16x8*)(&k_lds[buf][16 * h + row16][32 * c + 8 * grp]);
-            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[c], acc, 0, 0, 0);
-          }
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            float s = acc[r] * scale;
-            if (causal) {
-              const int key_g = j0 + 16 * h + 4 * grp + r;
-              const int q_g = i0 + row16;
-              if (key_g > q_g) s = -1e30f;
-            }
-            sv[h * 4 + r] = s;
-          }
-        }
-
-        // ---- online softmax (per q = row16; reduce across grp groups) ----
-        float mt = sv[0];
-#pragma unroll
-        for (int x = 1; x < 8; ++x) mt = fmaxf(mt, sv[x]);
-        mt = fmaxf(mt, __shfl_xor(mt, 16, kWave));
-        mt = fmaxf(mt, __shfl_xor(mt, 32, kWave));
-        const float m_new = fmaxf(m_run, mt);
-        const float alpha = __expf(m_run - m_new); // m_run=-1e30 -> exp(-inf)=0
-
-        float ps = 0.0f;
-#pragma unroll
-        for (int x = 0; x < 8; ++x) {
-          sv[x] = __expf(sv[x] - m_new);
-          ps += sv[x];
-        }
-        ps += __shfl_xor(ps, 16, kWave);
-        ps += __shfl_xor(ps, 32, kWave);
-        s_run = s_run * alpha + ps;
-        m_run = m_new;
-
-        // ---- P^T -> per-wave LDS slice (same-wave visibility only) ----
-        // 4 consecutive keys pack into one 8-byte write per half
-        typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
-#pragma unroll
-        for (int h = 0; h < 2; ++h) {
-          bf16x4 pw;
-#pragma unroll
-          for (int r = 0; r < 4; ++r) pw[r] = (__bf16)sv[h * 4 + r];
-          *(bf16x4*)(&p_lds[row16][16 * h + 4 * grp]) = pw;
-        }
-
-        // rescale rows (q_o = 4*grp + r) once; lane q_o holds alpha[q_o]
-        float a_o[4];
-#pragma unroll
-        for (int r = 0; r < 4; ++r) a_o[r] = __shfl(alpha, 4 * grp + r, kWave);
-
-        // ---- PV: A = P[q][key] from LDS, B = V[key][d] from LDS ----
-        const bf16x8 pf = *(const bf16x8*)(&p_lds[row16][8 * grp]);
+        // shared V fragments for both q sub-tiles (the amortization win)
+        bf16x8 vf[4];
 #pragma unroll
         for (int db = 0; db < 4; ++db) {
-          bf16x8 vf;
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            vf[j] = *(const __bf16*)(&v_lds[buf][8 * grp + j][16 * db + row16]);
+            vf[db][j] = *(const __bf16*)(&v_lds[buf][8 * grp + j][16 * db + row16]);
           }
+        }
+        typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
+
 #pragma unroll
-          for (int r = 0; r < 4; ++r) o_acc[db][r] *= a_o[r];
-          o_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, o_acc[db], 0, 0, 0);
+        for (int sub = 0; sub < 2; ++sub) {
+          const int qi0 = i0 + 16 * sub;
+          if (causal && j0 >= qi0 + kQT) continue; // tile fully past diagonal
+
+          // ---- S^T = K Q^T for two 16-key halves ----
+          float sv[8]; // h*4 + r: key = j0 + 16h + 4grp + r, query = row16
+#pragma unroll
+          for (int h = 0; h < 2; ++h) {
+            f32x4 acc = {0, 0, 0, 0};
+#pragma unroll
+            for (int c = 0; c < 2; ++c) {
+              const bf16x8 kf = *(const bf16x8*)(&k_lds[buf][16 * h + row16][32 * c + 8 * grp]);
+              acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[sub][c], acc, 0, 0, 0);
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              float s = acc[r] * scale;
+              if (causal) {
+                const int key_g = j0 + 16 * h + 4 * grp + r;
+                const int q_g = qi0 + row16;
+                if (key_g > q_g) s = -1e30f;
+              }
+              sv[h * 4 + r] = s;
+            }
+          }
+
+          // ---- online softmax (per q = row16; reduce across grp groups) ----
+          float mt = sv[0];
+#pragma unroll
+          for (int x = 1; x < 8; ++x) mt = fmaxf(mt, sv[x]);
+          mt = fmaxf(mt, __shfl_xor(mt, 16, kWave));
+          mt = fmaxf(mt, __shfl_xor(mt, 32, kWave));
+          const float m_new = fmaxf(m_run[sub], mt);
+          const float alpha = __expf(m_run[sub] - m_new); // -1e30 -> 0
+
+          float ps = 0.0f;
+#pragma unroll
+          for (int x = 0; x < 8; ++x) {
+            sv[x] = __expf(sv[x] - m_new);
+            ps += sv[x];
+          }
+          ps += __shfl_xor(ps, 16, kWave);
+          ps += __shfl_xor(ps, 32, kWave);
+          s_run[sub] = s_run[sub] * alpha + ps;
+          m_run[sub] = m_new;
+
+          // ---- P^T -> per-wave LDS slice (4 keys per 8-byte write) ----
+#pragma unroll
+          for (int h = 0; h < 2; ++h) {
+            bf16x4 pw;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) pw[r] = (__bf16)sv[h * 4 + r];
+            *(bf16x4*)(&p_lds[row16][16 * h + 4 * grp]) = pw;
+          }
+
+          float a_o[4];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) a_o[r] = __shfl(alpha, 4 * grp + r, kWave);
+
+          // ---- PV: A = P[q][key] from LDS, B = shared V fragments ----
+          const bf16x8 pf = *(const bf16x8*)(&p_lds[row16][8 * grp]);
+#pragma unroll
+          for (int db = 0; db < 4; ++db) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r) o_acc[sub][db][r] *= a_o[r];
+            o_acc[sub][db] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf[db], o_acc[sub][db], 0, 0, 0);
+          }
         }
       }
       if (has_next) {
@@ -193,18 +212,21 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     if (valid) {
       // ---- epilogue: O /= sumexp, store; lse = m + log(sumexp) ----
 #pragma unroll
-      for (int db = 0; db < 4; ++db) {
+      for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int q_o = 4 * grp + r;
-          const float denom = __shfl(s_run, q_o, kWave);
-          const float val = o_acc[db][r] / denom;
-          o[(int64_t)bh * N * kAttnD + (int64_t)(i0 + q_o) * kAttnD + 16 * db + row16] =
-              __float2bfloat16(val);
+        for (int db = 0; db < 4; ++db) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int q_o = 4 * grp + r;
+            const float denom = __shfl(s_run[sub], q_o, kWave);
+            const float val = o_acc[sub][db][r] / denom;
+            o[(int64_t)bh * N * kAttnD + (int64_t)(i0 + 16 * sub + q_o) * kAttnD + 16 * db +
+              row16] = __float2bfloat16(val);
+          }
         }
-      }
-      if (lane < 16) {
-        lse[(int64_t)bh * N + i0 + row16] = m_run + __logf(s_run);
+        if (lane < 16) {
+          lse[(int64_t)bh * N + i0 + 16 * sub + row16] = m_run[sub] + __logf(s_run[sub]);
+        }
       }
     }
   }
@@ -218,7 +240,7 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor
   const int B = q.size(0), H = q.size(1), N = q.size(2);
   TORCH_CHECK(N % kKT == 0, "N must be a multiple of 32");
   const int BH = B * H;
-  const int qrows_per_block = kWavesPerBlock * kQT;
+  const int qrows_per_block = kWavesPerBlock * 2 * kQT;
   const int64_t total_blocks = (int64_t)BH * ((N + qrows_per_block - 1) / qrows_per_block);
   const int blocks = (int)std::min<int64_t>(total_blocks, kMaxGrid);
   auto stream = c10::hip::getCurrentHIPStream();
