@@ -26,6 +26,7 @@
 namespace dmlamd {
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 constexpr int kAttnD = 64; // head dim (v1: fixed)
@@ -129,8 +130,6 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
             vf[db][j] = *(const __bf16*)(&v_lds[buf][8 * grp + j][16 * db + row16]);
           }
         }
-        typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
-
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
           const int qi0 = i0 + 16 * sub;
@@ -248,6 +247,343 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor
                      (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                      (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)o.data_ptr(),
                      lse.data_ptr<float>(), BH, N, (float)scale, causal);
+}
+
+// ===========================================================================
+// Backward (flash recomputation, ops/_attention_ref.py::flash_attn_bwd_tiled)
+// ===========================================================================
+//
+//   delta = rowsum(dO * O)                    (attn_bwd_delta_kernel)
+//   P^T   = exp(K Q^T * scale - lse[q])       (recomputed per tile pair)
+//   dP^T  = V dO^T
+//   dS^T  = P^T * (dP^T - delta[q]) * scale
+//   dV   += P^T  @ dO      dK += dS^T @ Q     (attn_bwd_dkdv_kernel,
+//                                              wave owns a 16-key tile)
+//   dQ   += dS @ K                            (attn_bwd_dq_kernel,
+//                                              wave owns a 16-q tile)
+//
+// Every MFMA reuses the forward's fragment patterns; the only transposes
+// (P^T/dS^T C-layout -> A operand) go through small per-wave LDS slices
+// exactly like the forward's P.
+
+// -------------------------------------------------- delta = rowsum(dO*O)
+__global__ void __launch_bounds__(kBlock) attn_bwd_delta_kernel(
+    const __hip_bfloat16* __restrict__ dout, const __hip_bfloat16* __restrict__ o,
+    float* __restrict__ delta, int64_t rows) {
+  // one wave per 8 rows: lane l -> row l/8, 8-elem chunk l%8
+  const int64_t stride = ((int64_t)gridDim.x * kBlock) / 8;
+  for (int64_t r0 = ((int64_t)blockIdx.x * kBlock + threadIdx.x) / 8; r0 < rows; r0 += stride) {
+    const int chunk = threadIdx.x & 7;
+    bf16x8 a = *(const bf16x8*)(dout + r0 * kAttnD + 8 * chunk);
+    bf16x8 b = *(const bf16x8*)(o + r0 * kAttnD + 8 * chunk);
+    float s = 0.0f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s += (float)a[j] * (float)b[j];
+    // reduce the 8 chunks of this row (lanes l..l+7 within the wave)
+#pragma unroll
+    for (int off = 4; off > 0; off >>= 1) s += __shfl_xor(s, off, kWave);
+    if (chunk == 0) delta[r0] = s;
+  }
+}
+
+// ------------------------------------------------------------- dK and dV
+// Block = 4 waves, each owning a 16-key tile of one (b,h); q-tiles of 32
+// rows (Q and dO staged in shared LDS) stream past. Per q-tile, per wave:
+// 2 MFMAs S^T, 2 MFMAs dP^T, 4+4 MFMAs dV/dK accumulation (k = 32 q rows).
+__global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv, int BH, int N,
+    float scale, bool causal) {
+  constexpr int kQStep = 32; // q rows per iteration (MFMA contraction width)
+  __shared__ __hip_bfloat16 q_lds[kQStep][kKVStride];
+  __shared__ __hip_bfloat16 do_lds[kQStep][kKVStride];
+  // per-wave transpose slices, [key16][q32(+skew)] rows for A-operand reads
+  __shared__ __hip_bfloat16 pt_lds_all[kWavesPerBlock][16][kPStride];
+  __shared__ __hip_bfloat16 dst_lds_all[kWavesPerBlock][16][kPStride];
+
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wave = threadIdx.x / kWave;
+  const int row16 = lane & 15;
+  const int grp = lane >> 4;
+  __hip_bfloat16(*pt_lds)[kPStride] = pt_lds_all[wave];
+  __hip_bfloat16(*dst_lds)[kPStride] = dst_lds_all[wave];
+
+  const int st_row = threadIdx.x >> 3; // 0..31
+  const int st_col = (threadIdx.x & 7) * 8;
+
+  const int keys_per_block = kWavesPerBlock * 16; // 64
+  const int nkb = (N + keys_per_block - 1) / keys_per_block;
+  const int64_t total_blocks = (int64_t)BH * nkb;
+
+  for (int64_t blk = blockIdx.x; blk < total_blocks; blk += gridDim.x) {
+    const int bh = blk / nkb;
+    const int kb0 = (blk - (int64_t)bh * nkb) * keys_per_block;
+    const int j0 = kb0 + wave * 16; // this wave's 16 keys
+    const bool valid = j0 < N;
+    const __hip_bfloat16* qp = q + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* kp = k + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* vp = v + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* dop = dout + (int64_t)bh * N * kAttnD;
+    const float* lsep = lse + (int64_t)bh * N;
+    const float* delp = delta + (int64_t)bh * N;
+
+    // wave-local K and V fragments for this key tile (row-major loads)
+    bf16x8 kf[2], vf2[2];
+    if (valid) {
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        kf[c] = *(const bf16x8*)(kp + (int64_t)(j0 + row16) * kAttnD + 32 * c + 8 * grp);
+        vf2[c] = *(const bf16x8*)(vp + (int64_t)(j0 + row16) * kAttnD + 32 * c + 8 * grp);
+      }
+    }
+
+    f32x4 dv_acc[4], dk_acc[4]; // rows = key (4grp+r), cols = 16*db + row16
+#pragma unroll
+    for (int db = 0; db < 4; ++db) {
+      dv_acc[db] = f32x4{0, 0, 0, 0};
+      dk_acc[db] = f32x4{0, 0, 0, 0};
+    }
+
+    const int i_start = causal ? (kb0 & ~(kQStep - 1)) : 0; // block-aligned diag
+    for (int i0 = i_start; i0 < N; i0 += kQStep) {
+      // ---- stage Q and dO (32 x 64 each; one bf16x8 per thread each) ----
+      *(bf16x8*)(&q_lds[st_row][st_col]) =
+          *(const bf16x8*)(qp + (int64_t)(i0 + st_row) * kAttnD + st_col);
+      *(bf16x8*)(&do_lds[st_row][st_col]) =
+          *(const bf16x8*)(dop + (int64_t)(i0 + st_row) * kAttnD + st_col);
+      __syncthreads();
+
+      if (valid && (!causal || i0 + kQStep > j0)) {
+        // two 16-q halves share this wave's key tile
+#pragma unroll
+        for (int hq = 0; hq < 2; ++hq) {
+          const int q0 = i0 + 16 * hq; // this half's first q row
+          const float lse_q = lsep[q0 + row16]; // lane's q = row16
+          const float del_q = delp[q0 + row16];
+
+          // ---- S^T = K Q^T (rows = key, cols = q) ----
+          f32x4 acc = {0, 0, 0, 0};
+#pragma unroll
+          for (int c = 0; c < 2; ++c) {
+            const bf16x8 qf = *(const bf16x8*)(&q_lds[16 * hq + row16][32 * c + 8 * grp]);
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[c], qf, acc, 0, 0, 0);
+          }
+          // ---- dP^T = V dO^T (same geometry) ----
+          f32x4 dpt = {0, 0, 0, 0};
+#pragma unroll
+          for (int c = 0; c < 2; ++c) {
+            const bf16x8 dof = *(const bf16x8*)(&do_lds[16 * hq + row16][32 * c + 8 * grp]);
+            dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf2[c], dof, dpt, 0, 0, 0);
+          }
+
+          // ---- P^T and dS^T in C layout; write transposed slices ----
+          // lane holds keys 4grp+r (rows), q = row16 (col)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int key_g = j0 + 4 * grp + r;
+            const int q_g = q0 + row16;
+            float p = 0.0f;
+            if (!causal || key_g <= q_g) {
+              p = __expf(acc[r] * scale - lse_q);
+            }
+            const float ds = p * (dpt[r] - del_q) * scale;
+            // [key][q] slices: q column of this half = 16*hq + row16
+            pt_lds[4 * grp + r][16 * hq + row16] = __float2bfloat16(p);
+            dst_lds[4 * grp + r][16 * hq + row16] = __float2bfloat16(ds);
+          }
+        }
+        // same-wave LDS visibility; A-operand reads below
+
+        // ---- dV += P^T @ dO ; dK += dS^T @ Q  (contraction over 32 q) ----
+        const bf16x8 ptf = *(const bf16x8*)(&pt_lds[row16][8 * grp]);
+        const bf16x8 dstf = *(const bf16x8*)(&dst_lds[row16][8 * grp]);
+#pragma unroll
+        for (int db = 0; db < 4; ++db) {
+          bf16x8 dob, qb;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            dob[j] = *(const __bf16*)(&do_lds[8 * grp + j][16 * db + row16]);
+            qb[j] = *(const __bf16*)(&q_lds[8 * grp + j][16 * db + row16]);
+          }
+          dv_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf, dob, dv_acc[db], 0, 0, 0);
+          dk_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf, qb, dk_acc[db], 0, 0, 0);
+        }
+      }
+      __syncthreads(); // staged tiles consumed before restage
+    }
+
+    if (valid) {
+#pragma unroll
+      for (int db = 0; db < 4; ++db) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key_o = 4 * grp + r;
+          dv[(int64_t)bh * N * kAttnD + (int64_t)(j0 + key_o) * kAttnD + 16 * db + row16] =
+              __float2bfloat16(dv_acc[db][r]);
+          dk[(int64_t)bh * N * kAttnD + (int64_t)(j0 + key_o) * kAttnD + 16 * db + row16] =
+              __float2bfloat16(dk_acc[db][r]);
+        }
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------------- dQ
+// Mirror of the forward: block = 4 waves x one 16-q tile; 32-key KV tiles
+// (K and V staged shared). dq[q][d] += dS[q][key] @ K[key][d].
+__global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    __hip_bfloat16* __restrict__ dq, int BH, int N, float scale, bool causal) {
+  __shared__ __hip_bfloat16 k_lds[kKT][kKVStride];
+  __shared__ __hip_bfloat16 v_lds[kKT][kKVStride];
+  __shared__ __hip_bfloat16 ds_lds_all[kWavesPerBlock][kQT][kPStride]; // [q][key32]
+
+  const int lane = threadIdx.x & (kWave - 1);
+  const int wave = threadIdx.x / kWave;
+  const int row16 = lane & 15;
+  const int grp = lane >> 4;
+  __hip_bfloat16(*ds_lds)[kPStride] = ds_lds_all[wave];
+
+  const int st_row = threadIdx.x >> 3;
+  const int st_col = (threadIdx.x & 7) * 8;
+
+  const int qrows_per_block = kWavesPerBlock * kQT; // 64
+  const int nqb = (N + qrows_per_block - 1) / qrows_per_block;
+  const int64_t total_blocks = (int64_t)BH * nqb;
+
+  for (int64_t blk = blockIdx.x; blk < total_blocks; blk += gridDim.x) {
+    const int bh = blk / nqb;
+    const int qb0 = (blk - (int64_t)bh * nqb) * qrows_per_block;
+    const int i0 = qb0 + wave * kQT;
+    const bool valid = i0 < N;
+    const __hip_bfloat16* qp = q + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* kp = k + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* vp = v + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* dop = dout + (int64_t)bh * N * kAttnD;
+    const float* lsep = lse + (int64_t)bh * N;
+    const float* delp = delta + (int64_t)bh * N;
+
+    bf16x8 qf[2], dof[2];
+    float lse_q = 0.0f, del_q = 0.0f;
+    if (valid) {
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        qf[c] = *(const bf16x8*)(qp + (int64_t)(i0 + row16) * kAttnD + 32 * c + 8 * grp);
+        dof[c] = *(const bf16x8*)(dop + (int64_t)(i0 + row16) * kAttnD + 32 * c + 8 * grp);
+      }
+      lse_q = lsep[i0 + row16];
+      del_q = delp[i0 + row16];
+    }
+
+    f32x4 dq_acc[4];
+#pragma unroll
+    for (int db = 0; db < 4; ++db) dq_acc[db] = f32x4{0, 0, 0, 0};
+
+    const int kv_end_block = causal ? min(qb0 + qrows_per_block, N) : N;
+    const int my_kv_end = causal ? (i0 + kQT) : N;
+
+    for (int j0 = 0; j0 < kv_end_block; j0 += kKT) {
+      *(bf16x8*)(&k_lds[st_row][st_col]) =
+          *(const bf16x8*)(kp + (int64_t)(j0 + st_row) * kAttnD + st_col);
+      *(bf16x8*)(&v_lds[st_row][st_col]) =
+          *(const bf16x8*)(vp + (int64_t)(j0 + st_row) * kAttnD + st_col);
+      __syncthreads();
+
+      if (valid && j0 < my_kv_end) {
+#pragma unroll
+        for (int h = 0; h < 2; ++h) { // two 16-key halves
+          // S^T rows = key (4grp+r), col = q (row16)
+          f32x4 acc = {0, 0, 0, 0};
+          f32x4 dpt = {0, 0, 0, 0};
+#pragma unroll
+          for (int c = 0; c < 2; ++c) {
+            const bf16x8 kfr = *(const bf16x8*)(&k_lds[16 * h + row16][32 * c + 8 * grp]);
+            const bf16x8 vfr = *(const bf16x8*)(&v_lds[16 * h + row16][32 * c + 8 * grp]);
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfr, qf[c], acc, 0, 0, 0);
+            dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfr, dof[c], dpt, 0, 0, 0);
+          }
+          // dS^T -> [q][key] slice (4 consecutive keys pack per write)
+          bf16x4 dsw;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int key_g = j0 + 16 * h + 4 * grp + r;
+            const int q_g = i0 + row16;
+            float p = 0.0f;
+            if (!causal || key_g <= q_g) {
+              p = __expf(acc[r] * scale - lse_q);
+            }
+            dsw[r] = (__bf16)(p * (dpt[r] - del_q) * scale);
+          }
+          *(bf16x4*)(&ds_lds[row16][16 * h + 4 * grp]) = dsw;
+        }
+
+        // ---- dq += dS @ K (contraction over the 32 keys) ----
+        const bf16x8 dsf = *(const bf16x8*)(&ds_lds[row16][8 * grp]);
+#pragma unroll
+        for (int db = 0; db < 4; ++db) {
+          bf16x8 kb;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            kb[j] = *(const __bf16*)(&k_lds[8 * grp + j][16 * db + row16]);
+          }
+          dq_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, kb, dq_acc[db], 0, 0, 0);
+        }
+      }
+      __syncthreads();
+    }
+
+    if (valid) {
+#pragma unroll
+      for (int db = 0; db < 4; ++db) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int q_o = 4 * grp + r;
+          dq[(int64_t)bh * N * kAttnD + (int64_t)(i0 + q_o) * kAttnD + 16 * db + row16] =
+              __float2bfloat16(dq_acc[db][r]);
+        }
+      }
+    }
+  }
+}
+
+void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout, at::Tensor o,
+              at::Tensor lse, at::Tensor dq, at::Tensor dk, at::Tensor dv, at::Tensor delta,
+              double scale, bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 && q.is_contiguous(),
+              "q must be contiguous bf16 [B,H,N,D]");
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == kAttnD, "bwd supports head dim 64");
+  const int B = q.size(0), H = q.size(1), N = q.size(2);
+  TORCH_CHECK(N % 64 == 0, "bwd requires N to be a multiple of 64");
+  const int BH = B * H;
+  TORCH_CHECK(delta.numel() >= (int64_t)BH * N && delta.scalar_type() == at::kFloat,
+              "delta workspace must be fp32[BH*N]");
+  auto stream = c10::hip::getCurrentHIPStream();
+
+  const int64_t rows = (int64_t)BH * N;
+  hipLaunchKernelGGL(attn_bwd_delta_kernel, dim3(grid_for(rows * 8, kBlock)), dim3(kBlock), 0,
+                     stream, (const __hip_bfloat16*)dout.data_ptr(),
+                     (const __hip_bfloat16*)o.data_ptr(), delta.data_ptr<float>(), rows);
+
+  const int64_t kv_blocks = (int64_t)BH * ((N + 63) / 64);
+  hipLaunchKernelGGL(attn_bwd_dkdv_kernel, dim3((int)std::min<int64_t>(kv_blocks, kMaxGrid)),
+                     dim3(kWavesPerBlock * kWave), 0, stream,
+                     (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                     (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(), BH, N,
+                     (float)scale, causal);
+
+  const int64_t q_blocks = (int64_t)BH * ((N + 63) / 64);
+  hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3((int)std::min<int64_t>(q_blocks, kMaxGrid)),
+                     dim3(kWavesPerBlock * kWave), 0, stream,
+                     (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
+                     (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     (__hip_bfloat16*)dq.data_ptr(), BH, N, (float)scale, causal);
 }
 
 } // namespace dmlamd

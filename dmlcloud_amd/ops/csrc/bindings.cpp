@@ -55,6 +55,9 @@ void ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Tensor sc
 // attention.hip (experimental)
 void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor lse,
               double scale, bool causal);
+void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout, at::Tensor o,
+              at::Tensor lse, at::Tensor dq, at::Tensor dk, at::Tensor dv, at::Tensor delta,
+              double scale, bool causal);
 
 } // namespace dmlamd
 
@@ -88,4 +91,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &dmlamd::ce_bwd, "Cross-entropy backward (dlogits in one pass)");
   m.def("attn_fwd", &dmlamd::attn_fwd,
         "EXPERIMENTAL flash-attention forward (MFMA bf16, D=64, emits lse)");
+  m.def("attn_bwd", &dmlamd::attn_bwd,
+        "EXPERIMENTAL flash-attention backward (recompute; dq/dk/dv)");
 }

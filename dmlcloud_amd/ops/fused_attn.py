@@ -1,0 +1,65 @@
+"""EXPERIMENTAL autograd wrapper for the MFMA flash-attention kernels.
+
+`sdpa(q, k, v, causal=True)` matches
+F.scaled_dot_product_attention(..., is_causal=causal) for bf16
+[B, H, N, 64] device tensors; anything else falls back to torch SDPA
+(AOTriton). Kept behind DMLCLOUD_FUSED_ATTN=1 for model use while the
+kernels trail AOTriton (profiles/attention_fwd_experimental.txt);
+numerics are verified in tests/test_gpu.py.
+"""
+
+import math
+import os
+
+import torch
+from torch.nn import functional as F
+
+from . import _C, is_available
+
+
+class _SdpaFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        B, H, N, D = q.shape
+        o = torch.empty_like(q)
+        lse = torch.empty(B, H, N, dtype=torch.float32, device=q.device)
+        _C.attn_fwd(q, k, v, o, lse, scale, causal)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, o, lse = ctx.saved_tensors
+        B, H, N, D = q.shape
+        dout = dout.contiguous()
+        dq = torch.empty_like(q)
+        dk = torch.empty_like(k)
+        dv = torch.empty_like(v)
+        delta = torch.empty(B * H * N, dtype=torch.float32, device=q.device)
+        _C.attn_bwd(q, k, v, dout, o, lse, dq, dk, dv, delta, ctx.scale, ctx.causal)
+        return dq, dk, dv, None, None
+
+
+def _usable(q):
+    return (
+        q.is_cuda
+        and is_available()
+        and q.dtype == torch.bfloat16
+        and q.dim() == 4
+        and q.shape[-1] == 64
+        and q.shape[-2] % 64 == 0
+    )
+
+
+def sdpa(q, k, v, causal: bool = True):
+    """Scaled dot-product attention; fused MFMA kernels when usable."""
+    if _usable(q):
+        scale = 1.0 / math.sqrt(q.shape[-1])
+        return _SdpaFn.apply(q.contiguous(), k.contiguous(), v.contiguous(), causal, scale)
+    return F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+
+
+def fused_attention_enabled() -> bool:
+    return os.environ.get('DMLCLOUD_FUSED_ATTN', '0') not in ('0', '', 'false')

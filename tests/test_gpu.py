@@ -484,6 +484,32 @@ class TestAttnFwdExperimental:
         )
         torch.testing.assert_close(o.float(), ref, rtol=3e-2, atol=3e-2)
 
+    @pytest.mark.parametrize('causal', [True, False])
+    @pytest.mark.parametrize('b,h,n', [(1, 1, 64), (2, 3, 128), (2, 4, 1024)])
+    def test_backward_matches_sdpa(self, causal, b, h, n):
+        from dmlcloud_amd.ops.fused_attn import sdpa
+
+        torch.manual_seed(3)
+        d = 64
+        q = (torch.randn(b, h, n, d, device=DEV) * 0.5).to(torch.bfloat16).requires_grad_(True)
+        k = (torch.randn(b, h, n, d, device=DEV) * 0.5).to(torch.bfloat16).requires_grad_(True)
+        v = (torch.randn(b, h, n, d, device=DEV) * 0.5).to(torch.bfloat16).requires_grad_(True)
+        do = (torch.randn(b, h, n, d, device=DEV) * 0.5).to(torch.bfloat16)
+
+        out = sdpa(q, k, v, causal=causal)
+        out.backward(do)
+
+        q2 = q.detach().float().requires_grad_(True)
+        k2 = k.detach().float().requires_grad_(True)
+        v2 = v.detach().float().requires_grad_(True)
+        ref = torch.nn.functional.scaled_dot_product_attention(q2, k2, v2, is_causal=causal)
+        ref.backward(do.float())
+
+        torch.testing.assert_close(out.float(), ref, rtol=3e-2, atol=3e-2)
+        torch.testing.assert_close(q.grad.float(), q2.grad, rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(k.grad.float(), k2.grad, rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(v.grad.float(), v2.grad, rtol=5e-2, atol=5e-2)
+
     def test_lse_matches_blueprint(self):
         import math
 
