@@ -89,6 +89,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
       }
     }
 
+    const float scale2 = scale * 1.44269504f; // fold log2(e): exp -> v_exp_f32
     float m_run[2] = {-1e30f, -1e30f};
     float s_run[2] = {0.0f, 0.0f};
     f32x4 o_acc[2][4];
@@ -147,7 +148,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-              float s = acc[r] * scale;
+              float s = acc[r] * scale2; // log2-domain scores
               if (causal) {
                 const int key_g = j0 + 16 * h + 4 * grp + r;
                 const int q_g = qi0 + row16;
@@ -164,12 +165,12 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
           mt = fmaxf(mt, __shfl_xor(mt, 16, kWave));
           mt = fmaxf(mt, __shfl_xor(mt, 32, kWave));
           const float m_new = fmaxf(m_run[sub], mt);
-          const float alpha = __expf(m_run[sub] - m_new); // -1e30 -> 0
+          const float alpha = __exp2f(m_run[sub] - m_new); // -1e30 -> 0
 
           float ps = 0.0f;
 #pragma unroll
           for (int x = 0; x < 8; ++x) {
-            sv[x] = __expf(sv[x] - m_new);
+            sv[x] = __exp2f(sv[x] - m_new); // raw v_exp_f32 rate
             ps += sv[x];
           }
           ps += __shfl_xor(ps, 16, kWave);
@@ -224,7 +225,9 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
           }
         }
         if (lane < 16) {
-          lse[(int64_t)bh * N + i0 + 16 * sub + row16] = m_run[sub] + __logf(s_run[sub]);
+          // convert the log2-domain stats back to natural-log lse
+          lse[(int64_t)bh * N + i0 + 16 * sub + row16] =
+              0.69314718f * m_run[sub] + __logf(s_run[sub]);
         }
       }
     }
@@ -360,7 +363,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
 #pragma unroll
         for (int hq = 0; hq < 2; ++hq) {
           const int q0 = i0 + 16 * hq; // this half's first q row
-          const float lse_q = lsep[q0 + row16]; // lane's q = row16
+          const float lse2_q = lsep[q0 + row16] * 1.44269504f; // log2 domain
           const float del_q = delp[q0 + row16];
 
           // ---- S^T = K Q^T (rows = key, cols = q) ----
@@ -386,7 +389,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
             const int q_g = q0 + row16;
             float p = 0.0f;
             if (!causal || key_g <= q_g) {
-              p = __expf(acc[r] * scale - lse_q);
+              p = __exp2f(acc[r] * (scale * 1.44269504f) - lse2_q);
             }
             const float ds = p * (dpt[r] - del_q) * scale;
             // [key][q] slices: q column of this half = 16*hq + row16
@@ -475,7 +478,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
         qf[c] = *(const bf16x8*)(qp + (int64_t)(i0 + row16) * kAttnD + 32 * c + 8 * grp);
         dof[c] = *(const bf16x8*)(dop + (int64_t)(i0 + row16) * kAttnD + 32 * c + 8 * grp);
       }
-      lse_q = lsep[i0 + row16];
+      lse_q = lsep[i0 + row16] * 1.44269504f; // log2 domain
       del_q = delp[i0 + row16];
     }
 
@@ -514,7 +517,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
             const int q_g = i0 + row16;
             float p = 0.0f;
             if (!causal || key_g <= q_g) {
-              p = __expf(acc[r] * scale - lse_q);
+              p = __exp2f(acc[r] * (scale * 1.44269504f) - lse_q); // lse_q in log2 domain
             }
             dsw[r] = (__bf16)(p * (dpt[r] - del_q) * scale);
           }
