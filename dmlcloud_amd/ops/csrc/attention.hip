@@ -165,12 +165,12 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
           mt = fmaxf(mt, __shfl_xor(mt, 16, kWave));
           mt = fmaxf(mt, __shfl_xor(mt, 32, kWave));
           const float m_new = fmaxf(m_run[sub], mt);
-          const float alpha = exp2f(m_run[sub] - m_new); // -1e30 -> 0
+          const float alpha = __builtin_amdgcn_exp2f(m_run[sub] - m_new); // -1e30 -> 0
 
           float ps = 0.0f;
 #pragma unroll
           for (int x = 0; x < 8; ++x) {
-            sv[x] = exp2f(sv[x] - m_new); // raw v_exp_f32 rate
+            sv[x] = __builtin_amdgcn_exp2f(sv[x] - m_new); // raw v_exp_f32 rate
             ps += sv[x];
           }
           ps += __shfl_xor(ps, 16, kWave);
@@ -389,7 +389,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
             const int q_g = q0 + row16;
             float p = 0.0f;
             if (!causal || key_g <= q_g) {
-              p = exp2f(acc[r] * (scale * 1.44269504f) - lse2_q);
+              p = __builtin_amdgcn_exp2f(acc[r] * (scale * 1.44269504f) - lse2_q);
             }
             const float ds = p * (dpt[r] - del_q) * scale;
             // [key][q] slices: q column of this half = 16*hq + row16
@@ -517,7 +517,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
             const int q_g = i0 + row16;
             float p = 0.0f;
             if (!causal || key_g <= q_g) {
-              p = exp2f(acc[r] * (scale * 1.44269504f) - lse_q); // lse_q in log2 domain
+              p = __builtin_amdgcn_exp2f(acc[r] * (scale * 1.44269504f) - lse_q); // lse_q in log2 domain
             }
             dsw[r] = (__bf16)(p * (dpt[r] - del_q) * scale);
           }
