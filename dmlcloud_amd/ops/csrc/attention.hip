@@ -52,7 +52,11 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     float* __restrict__ lse, int BH, int N, float scale, bool causal) {
   __shared__ __hip_bfloat16 p_lds_all[kWavesPerBlock][kQT][kPStride];
   __shared__ __hip_bfloat16 k_lds[2][kKT][kKVStride];
-  __shared__ __hip_bfloat16 v_lds[2][kKT][kKVStride];
+  // V lives in a TILED image read by ds_read_b64_tr_b16 (empirically
+  // probed semantics, tools/tr_probe.hip: 16 contiguous per-lane 8-byte
+  // addresses cover a 64-element region transposed as 4x16):
+  //   elem[(db*2+half)*256 + g*64 + jj*16 + c] = V[8g+4half+jj][16db+c]
+  __shared__ __hip_bfloat16 v_tr[2][2 * kKT * kAttnD / 2]; // [2][2048]
 
   const int lane = threadIdx.x & (kWave - 1);
   const int wave = threadIdx.x / kWave;
@@ -63,6 +67,9 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
   // staging coords: thread t loads row t/8, bf16x8 chunk t%8 of a 32x64 tile
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 8;
+  // tiled V image offset for this thread's (st_row, st_col) bf16x8 piece
+  const int st_vt = ((st_col >> 4) * 2 + ((st_row >> 2) & 1)) * 256 + (st_row >> 3) * 64 +
+                    (st_row & 3) * 16 + (st_col & 15);
 
   const int qrows_per_block = kWavesPerBlock * 2 * kQT; // 128 (2 sub-tiles per wave)
   const int nqb = (N + qrows_per_block - 1) / qrows_per_block;
@@ -106,8 +113,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     // ds_write targets the other buffer just before the single barrier.
     *(bf16x8*)(&k_lds[0][st_row][st_col]) =
         *(const bf16x8*)(kp + (int64_t)st_row * kAttnD + st_col);
-    *(bf16x8*)(&v_lds[0][st_row][st_col]) =
-        *(const bf16x8*)(vp + (int64_t)st_row * kAttnD + st_col);
+    *(bf16x8*)(&v_tr[0][st_vt]) = *(const bf16x8*)(vp + (int64_t)st_row * kAttnD + st_col);
     __syncthreads();
 
     const int ntiles = (kv_end_block + kKT - 1) / kKT;
@@ -122,14 +128,20 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
       }
 
       if (valid && j0 < my_kv_end) {
-        // shared V fragments for both q sub-tiles (the amortization win)
+        // shared V fragments for both q sub-tiles: two hardware transpose
+        // reads per fragment instead of 8 scalar reads + packing
+        typedef short short4v __attribute__((ext_vector_type(4)));
         bf16x8 vf[4];
 #pragma unroll
         for (int db = 0; db < 4; ++db) {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            vf[db][j] = *(const __bf16*)(&v_lds[buf][8 * grp + j][16 * db + row16]);
-          }
+          const short4v t0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) short4v*)&v_tr[buf][(db * 2 + 0) * 256 +
+                                                                     lane * 4]);
+          const short4v t1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (__attribute__((address_space(3))) short4v*)&v_tr[buf][(db * 2 + 1) * 256 +
+                                                                     lane * 4]);
+          __builtin_memcpy(&vf[db], &t0, 8);
+          __builtin_memcpy((char*)&vf[db] + 8, &t1, 8);
         }
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
@@ -204,7 +216,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
       }
       if (has_next) {
         *(bf16x8*)(&k_lds[buf ^ 1][st_row][st_col]) = knext;
-        *(bf16x8*)(&v_lds[buf ^ 1][st_row][st_col]) = vnext;
+        *(bf16x8*)(&v_tr[buf ^ 1][st_vt]) = vnext;
       }
       __syncthreads(); // readers of buf done AND buf^1 writes visible
     }
