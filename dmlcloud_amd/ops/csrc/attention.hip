@@ -39,6 +39,15 @@ constexpr int kWavesPerBlock = 4;
 // column offset span 16 distinct banks (rows at stride 88 bf16 = 44 dwords)
 constexpr int kKVStride = 88;
 
+// Tiled 32x64 LDS image index for ds_read_b64_tr_b16 consumption
+// (tools/tr_probe.hip): elem[(col/16*2 + row/4%2)*256 + row/8*64 +
+// row%4*16 + col%16] = T[row][col]. Row-major b128 reads also work on
+// the image (any 8 contiguous cols within a 16-col sub-tile row).
+__device__ __forceinline__ int vt_idx(int row, int col) {
+  return (((col >> 4) * 2 + ((row >> 2) & 1)) * 4 + (row >> 3)) * 64 + (row & 3) * 16 +
+         (col & 15);
+}
+
 // v2: the block's 4 waves own consecutive q-tiles of ONE (b,h) and share
 // cooperatively staged K/V LDS tiles (one bf16x8 global load per thread per
 // tile — coalesced — instead of per-lane scattered loads; v1 measured
@@ -312,8 +321,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
     __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv, int BH, int N,
     float scale, bool causal) {
   constexpr int kQStep = 32; // q rows per iteration (MFMA contraction width)
-  __shared__ __hip_bfloat16 q_lds[kQStep][kKVStride];
-  __shared__ __hip_bfloat16 do_lds[kQStep][kKVStride];
+  __shared__ __hip_bfloat16 q_lds[kQStep * kAttnD]; // tiled (vt_idx) image
+  __shared__ __hip_bfloat16 do_lds[kQStep * kAttnD]; // tiled (vt_idx) image
   // per-wave transpose slices, [key16][q32(+skew)] rows for A-operand reads
   __shared__ __hip_bfloat16 pt_lds_all[kWavesPerBlock][16][kPStride];
   __shared__ __hip_bfloat16 dst_lds_all[kWavesPerBlock][16][kPStride];
@@ -363,10 +372,10 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
 
     const int i_start = causal ? (kb0 & ~(kQStep - 1)) : 0; // block-aligned diag
     for (int i0 = i_start; i0 < N; i0 += kQStep) {
-      // ---- stage Q and dO (32 x 64 each; one bf16x8 per thread each) ----
-      *(bf16x8*)(&q_lds[st_row][st_col]) =
-          *(const bf16x8*)(qp + (int64_t)(i0 + st_row) * kAttnD + st_col);
-      *(bf16x8*)(&do_lds[st_row][st_col]) =
+      // ---- stage Q and dO into tiled images (one bf16x8 per thread) ----
+      const int st_t = vt_idx(st_row, st_col);
+      *(bf16x8*)(&q_lds[st_t]) = *(const bf16x8*)(qp + (int64_t)(i0 + st_row) * kAttnD + st_col);
+      *(bf16x8*)(&do_lds[st_t]) =
           *(const bf16x8*)(dop + (int64_t)(i0 + st_row) * kAttnD + st_col);
       __syncthreads();
 
@@ -380,16 +389,13 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
 
           // ---- S^T = K Q^T (rows = key, cols = q) ----
           f32x4 acc = {0, 0, 0, 0};
-#pragma unroll
-          for (int c = 0; c < 2; ++c) {
-            const bf16x8 qf = *(const bf16x8*)(&q_lds[16 * hq + row16][32 * c + 8 * grp]);
-            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[c], qf, acc, 0, 0, 0);
-          }
-          // ---- dP^T = V dO^T (same geometry) ----
           f32x4 dpt = {0, 0, 0, 0};
 #pragma unroll
           for (int c = 0; c < 2; ++c) {
-            const bf16x8 dof = *(const bf16x8*)(&do_lds[16 * hq + row16][32 * c + 8 * grp]);
+            const int rm = vt_idx(16 * hq + row16, 32 * c + 8 * grp);
+            const bf16x8 qfr = *(const bf16x8*)(&q_lds[rm]);
+            const bf16x8 dof = *(const bf16x8*)(&do_lds[rm]);
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[c], qfr, acc, 0, 0, 0);
             dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf2[c], dof, dpt, 0, 0, 0);
           }
 
@@ -414,13 +420,19 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
         // ---- dV += P^T @ dO ; dK += dS^T @ Q  (contraction over 32 q) ----
         const bf16x8 ptf = *(const bf16x8*)(&pt_lds[row16][8 * grp]);
         const bf16x8 dstf = *(const bf16x8*)(&dst_lds[row16][8 * grp]);
+        typedef short short4v __attribute__((ext_vector_type(4)));
 #pragma unroll
         for (int db = 0; db < 4; ++db) {
           bf16x8 dob, qb;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            dob[j] = *(const __bf16*)(&do_lds[8 * grp + j][16 * db + row16]);
-            qb[j] = *(const __bf16*)(&q_lds[8 * grp + j][16 * db + row16]);
+          for (int half = 0; half < 2; ++half) {
+            const int base = (db * 2 + half) * 256 + lane * 4;
+            const short4v td = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                (__attribute__((address_space(3))) short4v*)&do_lds[base]);
+            const short4v tq = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                (__attribute__((address_space(3))) short4v*)&q_lds[base]);
+            __builtin_memcpy((char*)&dob + 8 * half, &td, 8);
+            __builtin_memcpy((char*)&qb + 8 * half, &tq, 8);
           }
           dv_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ptf, dob, dv_acc[db], 0, 0, 0);
           dk_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dstf, qb, dk_acc[db], 0, 0, 0);
@@ -453,7 +465,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     __hip_bfloat16* __restrict__ dq, int BH, int N, float scale, bool causal) {
-  __shared__ __hip_bfloat16 k_lds[kKT][kKVStride];
+  __shared__ __hip_bfloat16 k_lds[kKT * kAttnD]; // tiled (vt_idx) image
   __shared__ __hip_bfloat16 v_lds[kKT][kKVStride];
   __shared__ __hip_bfloat16 ds_lds_all[kWavesPerBlock][kQT][kPStride]; // [q][key32]
 
@@ -502,7 +514,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
     const int my_kv_end = causal ? (i0 + kQT) : N;
 
     for (int j0 = 0; j0 < kv_end_block; j0 += kKT) {
-      *(bf16x8*)(&k_lds[st_row][st_col]) =
+      *(bf16x8*)(&k_lds[vt_idx(st_row, st_col)]) =
           *(const bf16x8*)(kp + (int64_t)(j0 + st_row) * kAttnD + st_col);
       *(bf16x8*)(&v_lds[st_row][st_col]) =
           *(const bf16x8*)(vp + (int64_t)(j0 + st_row) * kAttnD + st_col);
@@ -516,7 +528,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
           f32x4 dpt = {0, 0, 0, 0};
 #pragma unroll
           for (int c = 0; c < 2; ++c) {
-            const bf16x8 kfr = *(const bf16x8*)(&k_lds[16 * h + row16][32 * c + 8 * grp]);
+            const bf16x8 kfr = *(const bf16x8*)(&k_lds[vt_idx(16 * h + row16, 32 * c + 8 * grp)]);
             const bf16x8 vfr = *(const bf16x8*)(&v_lds[16 * h + row16][32 * c + 8 * grp]);
             acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfr, qf[c], acc, 0, 0, 0);
             dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfr, dof[c], dpt, 0, 0, 0);
@@ -538,12 +550,16 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
 
         // ---- dq += dS @ K (contraction over the 32 keys) ----
         const bf16x8 dsf = *(const bf16x8*)(&ds_lds[row16][8 * grp]);
+        typedef short short4v __attribute__((ext_vector_type(4)));
 #pragma unroll
         for (int db = 0; db < 4; ++db) {
           bf16x8 kb;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            kb[j] = *(const __bf16*)(&k_lds[8 * grp + j][16 * db + row16]);
+          for (int half = 0; half < 2; ++half) {
+            const short4v tk = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                (__attribute__((address_space(3))) short4v*)&k_lds[(db * 2 + half) * 256 +
+                                                                   lane * 4]);
+            __builtin_memcpy((char*)&kb + 8 * half, &tk, 8);
           }
           dq_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, kb, dq_acc[db], 0, 0, 0);
         }
