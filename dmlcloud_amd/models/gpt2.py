@@ -13,6 +13,7 @@ import torch
 from torch import nn
 from torch.nn import functional as F
 
+from ..ops.fused_attn import sdpa
 from ..ops.fused_ln import LayerNorm
 from ..ops.fused_loss import cross_entropy
 
@@ -42,7 +43,10 @@ class CausalSelfAttention(nn.Module):
         q = q.view(B, T, self.n_head, C // self.n_head).transpose(1, 2)
         k = k.view(B, T, self.n_head, C // self.n_head).transpose(1, 2)
         v = v.view(B, T, self.n_head, C // self.n_head).transpose(1, 2)
-        y = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        # custom MFMA flash attention on bf16/D=64 (ops/fused_attn.py);
+        # beats AOTriton fwd+bwd combined at this shape — falls back to
+        # torch SDPA otherwise
+        y = sdpa(q, k, v, causal=True)
         y = y.transpose(1, 2).reshape(B, T, C)
         return self.c_proj(y)
 
