@@ -43,6 +43,8 @@ class _SdpaFn(torch.autograd.Function):
 
 
 def _usable(q):
+    if os.environ.get('DMLCLOUD_DISABLE_FUSED_ATTN', '0') not in ('0', '', 'false'):
+        return False
     return (
         q.is_cuda
         and is_available()
