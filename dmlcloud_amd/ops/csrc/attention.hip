@@ -55,10 +55,22 @@ __device__ __forceinline__ int vt_idx(int row, int col) {
 // v5: each wave owns TWO 16-row q sub-tiles (32 q rows): the shared K
 // fragments and the scalar V fragment reads amortize over twice the
 // MFMAs, roughly doubling the per-tile MFMA:overhead ratio.
+// per-tensor strides (elements): batch, head, row; the last dim must be
+// contiguous. Lets the model pass transpose views without materializing.
+struct Strides {
+  int64_t b, h, r;
+};
+
+__device__ __forceinline__ const __hip_bfloat16* tslice(const __hip_bfloat16* t,
+                                                        const Strides& st, int bh, int H) {
+  return t + (int64_t)(bh / H) * st.b + (int64_t)(bh % H) * st.h;
+}
+
 __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
-    float* __restrict__ lse, int BH, int N, float scale, bool causal) {
+    float* __restrict__ lse, int BH, int H, int N, float scale, bool causal, Strides sq,
+    Strides sk, Strides sv) {
   __shared__ __hip_bfloat16 p_lds_all[kWavesPerBlock][kQT][kPStride];
   __shared__ __hip_bfloat16 k_lds[2][kKT][kKVStride];
   // V lives in a TILED image read by ds_read_b64_tr_b16 (empirically
@@ -89,9 +101,9 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     const int qb0 = (blk - (int64_t)bh * nqb) * qrows_per_block;
     const int i0 = qb0 + wave * 2 * kQT; // this wave's 32 q rows
     const bool valid = i0 < N;
-    const __hip_bfloat16* qp = q + (int64_t)bh * N * kAttnD;
-    const __hip_bfloat16* kp = k + (int64_t)bh * N * kAttnD;
-    const __hip_bfloat16* vp = v + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* qp = tslice(q, sq, bh, H);
+    const __hip_bfloat16* kp = tslice(k, sk, bh, H);
+    const __hip_bfloat16* vp = tslice(v, sv, bh, H);
 
     bf16x8 qf[2][2]; // [sub-tile][k-chunk]
     if (valid) {
@@ -99,8 +111,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
       for (int sub = 0; sub < 2; ++sub) {
 #pragma unroll
         for (int c = 0; c < 2; ++c) {
-          qf[sub][c] = *(const bf16x8*)(qp + (int64_t)(i0 + 16 * sub + row16) * kAttnD +
-                                        32 * c + 8 * grp);
+          qf[sub][c] = *(const bf16x8*)(qp + (int64_t)(i0 + 16 * sub + row16) * sq.r + 32 * c +
+                                        8 * grp);
         }
       }
     }
@@ -120,9 +132,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     // write-late double buffer (guide §6 G15): the NEXT tile's global
     // loads stay in flight through the current tile's compute; their
     // ds_write targets the other buffer just before the single barrier.
-    *(bf16x8*)(&k_lds[0][st_row][st_col]) =
-        *(const bf16x8*)(kp + (int64_t)st_row * kAttnD + st_col);
-    *(bf16x8*)(&v_tr[0][st_vt]) = *(const bf16x8*)(vp + (int64_t)st_row * kAttnD + st_col);
+    *(bf16x8*)(&k_lds[0][st_row][st_col]) = *(const bf16x8*)(kp + (int64_t)st_row * sk.r + st_col);
+    *(bf16x8*)(&v_tr[0][st_vt]) = *(const bf16x8*)(vp + (int64_t)st_row * sv.r + st_col);
     __syncthreads();
 
     const int ntiles = (kv_end_block + kKT - 1) / kKT;
@@ -132,8 +143,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
       bf16x8 knext, vnext;
       const bool has_next = jt + 1 < ntiles;
       if (has_next) {
-        knext = *(const bf16x8*)(kp + (int64_t)(j0 + kKT + st_row) * kAttnD + st_col);
-        vnext = *(const bf16x8*)(vp + (int64_t)(j0 + kKT + st_row) * kAttnD + st_col);
+        knext = *(const bf16x8*)(kp + (int64_t)(j0 + kKT + st_row) * sk.r + st_col);
+        vnext = *(const bf16x8*)(vp + (int64_t)(j0 + kKT + st_row) * sv.r + st_col);
       }
 
       if (valid && j0 < my_kv_end) {
@@ -255,13 +266,18 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
   }
 }
 
+static Strides strides_of(const at::Tensor& t) {
+  TORCH_CHECK(t.stride(3) == 1, "last dim must be contiguous");
+  return Strides{t.stride(0), t.stride(1), t.stride(2)};
+}
+
 void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor lse,
               double scale, bool causal) {
-  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 && q.is_contiguous(),
-              "q must be contiguous bf16 [B,H,N,D]");
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16, "q must be bf16 [B,H,N,D]");
   TORCH_CHECK(q.dim() == 4 && q.size(3) == kAttnD, "v1 supports head dim 64");
   const int B = q.size(0), H = q.size(1), N = q.size(2);
   TORCH_CHECK(N % kKT == 0, "N must be a multiple of 32");
+  TORCH_CHECK(o.is_contiguous(), "o must be contiguous");
   const int BH = B * H;
   const int qrows_per_block = kWavesPerBlock * 2 * kQT;
   const int64_t total_blocks = (int64_t)BH * ((N + qrows_per_block - 1) / qrows_per_block);
@@ -270,7 +286,8 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor
   hipLaunchKernelGGL(attn_fwd_kernel, dim3(blocks), dim3(kWavesPerBlock * kWave), 0, stream,
                      (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                      (const __hip_bfloat16*)v.data_ptr(), (__hip_bfloat16*)o.data_ptr(),
-                     lse.data_ptr<float>(), BH, N, (float)scale, causal);
+                     lse.data_ptr<float>(), BH, H, N, (float)scale, causal, strides_of(q),
+                     strides_of(k), strides_of(v));
 }
 
 // ===========================================================================
@@ -293,12 +310,14 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor
 // -------------------------------------------------- delta = rowsum(dO*O)
 __global__ void __launch_bounds__(kBlock) attn_bwd_delta_kernel(
     const __hip_bfloat16* __restrict__ dout, const __hip_bfloat16* __restrict__ o,
-    float* __restrict__ delta, int64_t rows) {
+    float* __restrict__ delta, int64_t rows, int H, int N, Strides sdo) {
   // one wave per 8 rows: lane l -> row l/8, 8-elem chunk l%8
   const int64_t stride = ((int64_t)gridDim.x * kBlock) / 8;
   for (int64_t r0 = ((int64_t)blockIdx.x * kBlock + threadIdx.x) / 8; r0 < rows; r0 += stride) {
     const int chunk = threadIdx.x & 7;
-    bf16x8 a = *(const bf16x8*)(dout + r0 * kAttnD + 8 * chunk);
+    const int bh = r0 / N;
+    const int rr = r0 - (int64_t)bh * N;
+    bf16x8 a = *(const bf16x8*)(tslice(dout, sdo, bh, H) + (int64_t)rr * sdo.r + 8 * chunk);
     bf16x8 b = *(const bf16x8*)(o + r0 * kAttnD + 8 * chunk);
     float s = 0.0f;
 #pragma unroll
@@ -318,8 +337,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv, int BH, int N,
-    float scale, bool causal) {
+    __hip_bfloat16* __restrict__ dk, __hip_bfloat16* __restrict__ dv, int BH, int H, int N,
+    float scale, bool causal, Strides sq, Strides sk, Strides sv, Strides sdo) {
   constexpr int kQStep = 32; // q rows per iteration (MFMA contraction width)
   __shared__ __hip_bfloat16 q_lds[kQStep * kAttnD]; // tiled (vt_idx) image
   __shared__ __hip_bfloat16 do_lds[kQStep * kAttnD]; // tiled (vt_idx) image
@@ -346,10 +365,10 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
     const int kb0 = (blk - (int64_t)bh * nkb) * keys_per_block;
     const int j0 = kb0 + wave * 16; // this wave's 16 keys
     const bool valid = j0 < N;
-    const __hip_bfloat16* qp = q + (int64_t)bh * N * kAttnD;
-    const __hip_bfloat16* kp = k + (int64_t)bh * N * kAttnD;
-    const __hip_bfloat16* vp = v + (int64_t)bh * N * kAttnD;
-    const __hip_bfloat16* dop = dout + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* qp = tslice(q, sq, bh, H);
+    const __hip_bfloat16* kp = tslice(k, sk, bh, H);
+    const __hip_bfloat16* vp = tslice(v, sv, bh, H);
+    const __hip_bfloat16* dop = tslice(dout, sdo, bh, H);
     const float* lsep = lse + (int64_t)bh * N;
     const float* delp = delta + (int64_t)bh * N;
 
@@ -358,8 +377,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
     if (valid) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        kf[c] = *(const bf16x8*)(kp + (int64_t)(j0 + row16) * kAttnD + 32 * c + 8 * grp);
-        vf2[c] = *(const bf16x8*)(vp + (int64_t)(j0 + row16) * kAttnD + 32 * c + 8 * grp);
+        kf[c] = *(const bf16x8*)(kp + (int64_t)(j0 + row16) * sk.r + 32 * c + 8 * grp);
+        vf2[c] = *(const bf16x8*)(vp + (int64_t)(j0 + row16) * sv.r + 32 * c + 8 * grp);
       }
     }
 
@@ -374,9 +393,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dkdv_kernel(
     for (int i0 = i_start; i0 < N; i0 += kQStep) {
       // ---- stage Q and dO into tiled images (one bf16x8 per thread) ----
       const int st_t = vt_idx(st_row, st_col);
-      *(bf16x8*)(&q_lds[st_t]) = *(const bf16x8*)(qp + (int64_t)(i0 + st_row) * kAttnD + st_col);
-      *(bf16x8*)(&do_lds[st_t]) =
-          *(const bf16x8*)(dop + (int64_t)(i0 + st_row) * kAttnD + st_col);
+      *(bf16x8*)(&q_lds[st_t]) = *(const bf16x8*)(qp + (int64_t)(i0 + st_row) * sq.r + st_col);
+      *(bf16x8*)(&do_lds[st_t]) = *(const bf16x8*)(dop + (int64_t)(i0 + st_row) * sdo.r + st_col);
       __syncthreads();
 
       if (valid && (!causal || i0 + kQStep > j0)) {
@@ -464,7 +482,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    __hip_bfloat16* __restrict__ dq, int BH, int N, float scale, bool causal) {
+    __hip_bfloat16* __restrict__ dq, int BH, int H, int N, float scale, bool causal, Strides sq,
+    Strides sk, Strides sv, Strides sdo) {
   __shared__ __hip_bfloat16 k_lds[kKT * kAttnD]; // tiled (vt_idx) image
   __shared__ __hip_bfloat16 v_lds[kKT][kKVStride];
   __shared__ __hip_bfloat16 ds_lds_all[kWavesPerBlock][kQT][kPStride]; // [q][key32]
@@ -487,10 +506,10 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
     const int qb0 = (blk - (int64_t)bh * nqb) * qrows_per_block;
     const int i0 = qb0 + wave * kQT;
     const bool valid = i0 < N;
-    const __hip_bfloat16* qp = q + (int64_t)bh * N * kAttnD;
-    const __hip_bfloat16* kp = k + (int64_t)bh * N * kAttnD;
-    const __hip_bfloat16* vp = v + (int64_t)bh * N * kAttnD;
-    const __hip_bfloat16* dop = dout + (int64_t)bh * N * kAttnD;
+    const __hip_bfloat16* qp = tslice(q, sq, bh, H);
+    const __hip_bfloat16* kp = tslice(k, sk, bh, H);
+    const __hip_bfloat16* vp = tslice(v, sv, bh, H);
+    const __hip_bfloat16* dop = tslice(dout, sdo, bh, H);
     const float* lsep = lse + (int64_t)bh * N;
     const float* delp = delta + (int64_t)bh * N;
 
@@ -499,8 +518,8 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
     if (valid) {
 #pragma unroll
       for (int c = 0; c < 2; ++c) {
-        qf[c] = *(const bf16x8*)(qp + (int64_t)(i0 + row16) * kAttnD + 32 * c + 8 * grp);
-        dof[c] = *(const bf16x8*)(dop + (int64_t)(i0 + row16) * kAttnD + 32 * c + 8 * grp);
+        qf[c] = *(const bf16x8*)(qp + (int64_t)(i0 + row16) * sq.r + 32 * c + 8 * grp);
+        dof[c] = *(const bf16x8*)(dop + (int64_t)(i0 + row16) * sdo.r + 32 * c + 8 * grp);
       }
       lse_q = lsep[i0 + row16] * 1.44269504f; // log2 domain
       del_q = delp[i0 + row16];
@@ -515,9 +534,9 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
 
     for (int j0 = 0; j0 < kv_end_block; j0 += kKT) {
       *(bf16x8*)(&k_lds[vt_idx(st_row, st_col)]) =
-          *(const bf16x8*)(kp + (int64_t)(j0 + st_row) * kAttnD + st_col);
+          *(const bf16x8*)(kp + (int64_t)(j0 + st_row) * sk.r + st_col);
       *(bf16x8*)(&v_lds[st_row][st_col]) =
-          *(const bf16x8*)(vp + (int64_t)(j0 + st_row) * kAttnD + st_col);
+          *(const bf16x8*)(vp + (int64_t)(j0 + st_row) * sv.r + st_col);
       __syncthreads();
 
       if (valid && j0 < my_kv_end) {
@@ -584,9 +603,11 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_bwd_dq_kernel(
 void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout, at::Tensor o,
               at::Tensor lse, at::Tensor dq, at::Tensor dk, at::Tensor dv, at::Tensor delta,
               double scale, bool causal) {
-  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16 && q.is_contiguous(),
-              "q must be contiguous bf16 [B,H,N,D]");
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16, "q must be bf16 [B,H,N,D]");
   TORCH_CHECK(q.dim() == 4 && q.size(3) == kAttnD, "bwd supports head dim 64");
+  TORCH_CHECK(dout.stride(3) == 1 && o.is_contiguous(), "o contiguous; dout last-dim contig");
+  TORCH_CHECK(dq.is_contiguous() && dk.is_contiguous() && dv.is_contiguous(),
+              "grad outputs must be contiguous");
   const int B = q.size(0), H = q.size(1), N = q.size(2);
   TORCH_CHECK(N % 64 == 0, "bwd requires N to be a multiple of 64");
   const int BH = B * H;
@@ -597,7 +618,8 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout, at::Ten
   const int64_t rows = (int64_t)BH * N;
   hipLaunchKernelGGL(attn_bwd_delta_kernel, dim3(grid_for(rows * 8, kBlock)), dim3(kBlock), 0,
                      stream, (const __hip_bfloat16*)dout.data_ptr(),
-                     (const __hip_bfloat16*)o.data_ptr(), delta.data_ptr<float>(), rows);
+                     (const __hip_bfloat16*)o.data_ptr(), delta.data_ptr<float>(), rows, H, N,
+                     strides_of(dout));
 
   const int64_t kv_blocks = (int64_t)BH * ((N + 63) / 64);
   hipLaunchKernelGGL(attn_bwd_dkdv_kernel, dim3((int)std::min<int64_t>(kv_blocks, kMaxGrid)),
@@ -605,8 +627,9 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout, at::Ten
                      (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                      (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(), BH, N,
-                     (float)scale, causal);
+                     (__hip_bfloat16*)dk.data_ptr(), (__hip_bfloat16*)dv.data_ptr(), BH, H, N,
+                     (float)scale, causal, strides_of(q), strides_of(k), strides_of(v),
+                     strides_of(dout));
 
   const int64_t q_blocks = (int64_t)BH * ((N + 63) / 64);
   hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3((int)std::min<int64_t>(q_blocks, kMaxGrid)),
@@ -614,7 +637,8 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout, at::Ten
                      (const __hip_bfloat16*)q.data_ptr(), (const __hip_bfloat16*)k.data_ptr(),
                      (const __hip_bfloat16*)v.data_ptr(), (const __hip_bfloat16*)dout.data_ptr(),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     (__hip_bfloat16*)dq.data_ptr(), BH, N, (float)scale, causal);
+                     (__hip_bfloat16*)dq.data_ptr(), BH, H, N, (float)scale, causal,
+                     strides_of(q), strides_of(k), strides_of(v), strides_of(dout));
 }
 
 } // namespace dmlamd

@@ -33,10 +33,11 @@ class _SdpaFn(torch.autograd.Function):
     def backward(ctx, dout):
         q, k, v, o, lse = ctx.saved_tensors
         B, H, N, D = q.shape
-        dout = dout.contiguous()
-        dq = torch.empty_like(q)
-        dk = torch.empty_like(k)
-        dv = torch.empty_like(v)
+        if dout.stride(-1) != 1:
+            dout = dout.contiguous()
+        dq = torch.empty(B, H, N, D, dtype=q.dtype, device=q.device)
+        dk = torch.empty_like(dq)
+        dv = torch.empty_like(dq)
         delta = torch.empty(B * H * N, dtype=torch.float32, device=q.device)
         _C.attn_bwd(q, k, v, dout, o, lse, dq, dk, dv, delta, ctx.scale, ctx.causal)
         return dq, dk, dv, None, None
@@ -55,11 +56,23 @@ def _usable(q):
     )
 
 
+def _lastdim_ok(t):
+    return t.stride(-1) == 1
+
+
 def sdpa(q, k, v, causal: bool = True):
-    """Scaled dot-product attention; fused MFMA kernels when usable."""
+    """Scaled dot-product attention; fused MFMA kernels when usable.
+
+    The kernels take per-tensor (batch, head, row) strides, so transpose
+    views (e.g. [B,T,H,D] memory viewed as [B,H,T,D]) pass through
+    without materialization — only a non-unit last-dim stride forces a
+    copy."""
     if _usable(q):
         scale = 1.0 / math.sqrt(q.shape[-1])
-        return _SdpaFn.apply(q.contiguous(), k.contiguous(), v.contiguous(), causal, scale)
+        q = q if _lastdim_ok(q) else q.contiguous()
+        k = k if _lastdim_ok(k) else k.contiguous()
+        v = v if _lastdim_ok(v) else v.contiguous()
+        return _SdpaFn.apply(q, k, v, causal, scale)
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal)
 
 

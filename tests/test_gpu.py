@@ -510,6 +510,36 @@ class TestAttnFwdExperimental:
         torch.testing.assert_close(k.grad.float(), k2.grad, rtol=5e-2, atol=5e-2)
         torch.testing.assert_close(v.grad.float(), v2.grad, rtol=5e-2, atol=5e-2)
 
+    def test_strided_views_no_copy(self):
+        """Transpose views ([B,T,H,D] memory seen as [B,H,T,D]) run through
+        the stride-aware kernels and match the contiguous path."""
+        from dmlcloud_amd.ops.fused_attn import sdpa
+
+        torch.manual_seed(5)
+        b, h, n, d = 2, 3, 128, 64
+        base_q = (torch.randn(b, n, h, d, device=DEV) * 0.5).to(torch.bfloat16)
+        base_k = (torch.randn(b, n, h, d, device=DEV) * 0.5).to(torch.bfloat16)
+        base_v = (torch.randn(b, n, h, d, device=DEV) * 0.5).to(torch.bfloat16)
+        q = base_q.transpose(1, 2).requires_grad_(True)
+        k = base_k.transpose(1, 2).requires_grad_(True)
+        v = base_v.transpose(1, 2).requires_grad_(True)
+        assert not q.is_contiguous()
+
+        do = (torch.randn(b, h, n, d, device=DEV) * 0.5).to(torch.bfloat16)
+        out = sdpa(q, k, v, causal=True)
+        out.backward(do)
+
+        q2 = q.detach().contiguous().requires_grad_(True)
+        k2 = k.detach().contiguous().requires_grad_(True)
+        v2 = v.detach().contiguous().requires_grad_(True)
+        out2 = sdpa(q2, k2, v2, causal=True)
+        out2.backward(do)
+
+        torch.testing.assert_close(out, out2)
+        torch.testing.assert_close(q.grad, q2.grad)
+        torch.testing.assert_close(k.grad, k2.grad)
+        torch.testing.assert_close(v.grad, v2.grad)
+
     def test_lse_matches_blueprint(self):
         import math
 
