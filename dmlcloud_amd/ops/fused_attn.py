@@ -21,7 +21,8 @@ class _SdpaFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
         B, H, N, D = q.shape
-        o = torch.empty_like(q)
+        # empty_like would inherit a transpose-view's strides; o is ours
+        o = torch.empty(B, H, N, D, dtype=q.dtype, device=q.device)
         lse = torch.empty(B, H, N, dtype=torch.float32, device=q.device)
         _C.attn_fwd(q, k, v, o, lse, scale, causal)
         ctx.save_for_backward(q, k, v, o, lse)
