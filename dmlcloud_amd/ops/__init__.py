@@ -191,7 +191,10 @@ def _norm_ws(device):
 
 
 def l2_norm(flat: torch.Tensor) -> torch.Tensor:
-    """Deterministic L2 norm of a flat fp32 buffer; returns fp32[1] on device."""
+    """Deterministic L2 norm of a flat fp32/bf16 buffer; returns fp32[1]
+    on device. The result is a view of a per-device scratch buffer
+    (stable pointers for hipGraph capture): clone it if you need the
+    value to survive a later l2_norm/clip_grad_norm_ call."""
     if flat.is_cuda:
         _require_ext()
         partials, out = _norm_ws(flat.device)

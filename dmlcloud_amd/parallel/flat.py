@@ -89,7 +89,9 @@ class FlatReplica:
         in backward-completion order over contiguous flat segments) and
         grad_sync() just waits — overlapping communication with the rest
         of backward like DDP does, but over the flat buffer. Leave None
-        for the single-collective mode (hipGraph-capturable)."""
+        for the single-collective mode, which is the one to use with
+        GraphedStep: hook-driven async collectives rely on host-side
+        bookkeeping that a replayed hipGraph would not re-execute."""
         if dtype not in (torch.float32, torch.bfloat16):
             raise ValueError('FlatReplica supports fp32 or bf16 parameter buffers')
         self.module = module
