@@ -62,6 +62,66 @@ def test_chunks_within_bounds(n, chunk, world, overlap):
             assert end - start == chunk + overlap
 
 
+@settings(max_examples=60, deadline=None)
+@given(
+    shapes=st.sampled_from([(3,), (2, 5), (4, 1), (2, 3, 2)]),
+    n=st.integers(1, 12),
+    op=st.sampled_from(['MEAN', 'SUM', 'MIN', 'MAX']),
+    seed=st.integers(0, 1000),
+)
+def test_reducer_matches_stack_reference(shapes, n, op, seed):
+    """Property: the O(1) accumulator reducer == the reference's
+    torch.stack + reduce over [0] + dims for any value sequence."""
+    import torch
+
+    from dmlcloud_amd.metrics import MetricReducer, Reduction
+
+    torch.manual_seed(seed)
+    values = [torch.randn(*shapes) for _ in range(n)]
+    reduction = Reduction(op)
+    r = MetricReducer(reduction=reduction)
+    r.extend(values)
+    out = r.reduce_locally()
+
+    stacked = torch.stack(values)
+    expected = {
+        'MEAN': stacked.mean(),
+        'SUM': stacked.sum(),
+        'MIN': stacked.amin(),
+        'MAX': stacked.amax(),
+    }[op]
+    torch.testing.assert_close(out.to(expected.dtype), expected, rtol=1e-5, atol=1e-5)
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    n=st.integers(1, 8),
+    dim=st.sampled_from([[0], [1], [0, 1]]),
+    op=st.sampled_from(['SUM', 'MIN', 'MAX', 'MEAN']),
+    seed=st.integers(0, 100),
+)
+def test_reducer_partial_dims_property(n, dim, op, seed):
+    import torch
+
+    from dmlcloud_amd.metrics import MetricReducer, Reduction
+
+    torch.manual_seed(seed)
+    values = [torch.randn(3, 4) for _ in range(n)]
+    r = MetricReducer(reduction=Reduction(op), dim=dim)
+    r.extend(values)
+    out = r.reduce_locally()
+
+    stacked = torch.stack(values)
+    dims = [0] + [d + 1 for d in dim]
+    expected = {
+        'MEAN': stacked.mean(dims),
+        'SUM': stacked.sum(dims),
+        'MIN': stacked.amin(dims),
+        'MAX': stacked.amax(dims),
+    }[op]
+    torch.testing.assert_close(out.to(expected.dtype), expected, rtol=1e-5, atol=1e-5)
+
+
 @settings(max_examples=100, deadline=None)
 @given(
     data=st.dictionaries(
