@@ -202,7 +202,11 @@ class FlatReplica:
     def load_state_dict(self, *args, **kwargs):
         res = self.module.load_state_dict(*args, **kwargs)
         # load_state_dict writes THROUGH the views (copy_ semantics), so the
-        # flat buffer stays authoritative; nothing else to do.
+        # flat buffer stays authoritative. Refresh the fp32 master from the
+        # loaded params (best available info; an optimizer load_state_dict
+        # afterwards restores the exact master).
+        if self.flat_master is not None:
+            self.flat_master.copy_(self.flat_param.to(torch.float32))
         return res
 
     def parameters(self):

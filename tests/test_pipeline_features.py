@@ -85,6 +85,40 @@ class TestCheckpointIntegration:
         assert p3.resumed is False
 
 
+class TestFlatCheckpointRoundtrip:
+    def test_bf16_flat_state_roundtrip(self, torch_distributed, tmp_path):
+        """bf16 flat replica + FlatAdam full-state save/load restores the
+        fp32 master exactly."""
+        from dmlcloud_amd.checkpoint import load_tensor_state, save_tensor_state
+        from dmlcloud_amd.parallel import FlatAdam, FlatReplica
+
+        torch.manual_seed(0)
+        model = torch.nn.Linear(10, 6)
+        replica = FlatReplica(model, dtype=torch.bfloat16)
+        opt = FlatAdam(replica, lr=1e-2)
+        x = torch.randn(4, 10).to(torch.bfloat16)
+        for _ in range(2):
+            replica.zero_grad()
+            replica(x).float().pow(2).mean().backward()
+            opt.step()
+
+        state = {'model': replica.state_dict(), 'opt': opt.state_dict()}
+        path = tmp_path / 's.dmlt'
+        save_tensor_state(state, path)
+        loaded = load_tensor_state(path)
+
+        model2 = torch.nn.Linear(10, 6)
+        replica2 = FlatReplica(model2, dtype=torch.bfloat16)
+        opt2 = FlatAdam(replica2, lr=1e-2)
+        replica2.load_state_dict(loaded['model'])
+        opt2.load_state_dict(loaded['opt'])
+
+        torch.testing.assert_close(replica2.flat_master, replica.flat_master)
+        torch.testing.assert_close(replica2.flat_param, replica.flat_param)
+        torch.testing.assert_close(opt2.exp_avg, opt.exp_avg)
+        assert opt2.step_t.item() == 2
+
+
 class TestGraphedStepCpu:
     def test_eager_fallback(self):
         from dmlcloud_amd.parallel import GraphedStep
