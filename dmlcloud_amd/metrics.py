@@ -345,20 +345,32 @@ class MetricTracker:
 
         Returns {name: all_empty_bool}; raises on rank divergence with the
         reference's message (reference metrics.py:127-128).
+
+        The vote also carries the metric NAME LIST: the fused all_reduce
+        in reduce_all assigns reduced values by registration order, so
+        rank-divergent registration order would silently mis-assign
+        values. Any mismatch in names or their order raises here instead.
         """
         global_names = [n for n in names if self.reducers[n].globally]
         result = {}
         if not global_names:
             return result
-        local = [self.reducers[n]._count is None for n in global_names]
+        local = (global_names, [self.reducers[n]._count is None for n in global_names])
         world = dist.get_world_size() if dist.is_initialized() else 1
         if world == 1:
             gathered = [local]
         else:
             gathered = [None] * world
             dist.all_gather_object(gathered, local)
+        for r, (rank_names, _) in enumerate(gathered):
+            if rank_names != global_names:
+                raise ValueError(
+                    f'Ranks disagree on the set/order of tracked metrics '
+                    f'(rank {dist.get_rank()}: {global_names} vs rank {r}: {rank_names}). '
+                    'Metrics must be registered in the same order on every rank.'
+                )
         for i, name in enumerate(global_names):
-            flags = [g[i] for g in gathered]
+            flags = [g[1][i] for g in gathered]
             if any(flags):
                 if len(flags) > 1 and not all(flags):
                     raise ValueError('Some workers tracked values this epoch and some did not. This is likely a bug.')

@@ -190,26 +190,34 @@ def _norm_ws(device):
     return ws
 
 
-def l2_norm(flat: torch.Tensor) -> torch.Tensor:
+def l2_norm(flat: torch.Tensor, norm_scale: float = 1.0) -> torch.Tensor:
     """Deterministic L2 norm of a flat fp32/bf16 buffer; returns fp32[1]
     on device. The result is a view of a per-device scratch buffer
     (stable pointers for hipGraph capture): clone it if you need the
-    value to survive a later l2_norm/clip_grad_norm_ call."""
+    value to survive a later l2_norm/clip_grad_norm_ call.
+
+    norm_scale pre-multiplies the norm: pass 1/world_size when `flat`
+    holds a rank-summed gradient to get the averaged-gradient norm."""
     if flat.is_cuda:
         _require_ext()
         partials, out = _norm_ws(flat.device)
-        _C.l2_norm_and_scale(flat, partials, out, -1.0, False)
+        _C.l2_norm_and_scale(flat, partials, out, -1.0, False, float(norm_scale))
         return out[:1]
-    return ref.l2_norm(flat)
+    return ref.l2_norm(flat) * norm_scale
 
 
-def clip_grad_norm_(flat: torch.Tensor, max_norm: float) -> torch.Tensor:
-    """Clip flat grads by global L2 norm in-place; returns norm (device, no sync)."""
+def clip_grad_norm_(flat: torch.Tensor, max_norm: float, norm_scale: float = 1.0) -> torch.Tensor:
+    """Clip flat grads by global L2 norm in-place; returns norm (device, no sync).
+
+    The clip threshold compares `||flat|| * norm_scale` against max_norm,
+    so a flat buffer holding the SUM of per-rank gradients clips with DDP
+    semantics when norm_scale = 1/world_size (the later 1/world averaging
+    in the fused optimizer then lands exactly on the clipped average)."""
     if flat.is_cuda:
         _require_ext()
         partials, out = _norm_ws(flat.device)
-        _C.l2_norm_and_scale(flat, partials, out, float(max_norm), True)
+        _C.l2_norm_and_scale(flat, partials, out, float(max_norm), True, float(norm_scale))
         return out[:1]
-    norm = ref.l2_norm(flat)
+    norm = ref.l2_norm(flat) * norm_scale
     ref.clip_by_norm_(flat, norm[0], max_norm)
     return norm

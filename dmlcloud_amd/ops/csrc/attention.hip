@@ -275,9 +275,15 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor
               double scale, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16, "q must be bf16 [B,H,N,D]");
   TORCH_CHECK(q.dim() == 4 && q.size(3) == kAttnD, "v1 supports head dim 64");
+  TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes(),
+              "self-attention geometry required: k/v shapes must equal q");
+  TORCH_CHECK(k.scalar_type() == at::kBFloat16 && v.scalar_type() == at::kBFloat16,
+              "k/v must be bf16");
   const int B = q.size(0), H = q.size(1), N = q.size(2);
   TORCH_CHECK(N % kKT == 0, "N must be a multiple of 32");
-  TORCH_CHECK(o.is_contiguous(), "o must be contiguous");
+  TORCH_CHECK(o.is_contiguous() && o.sizes() == q.sizes(), "o must be contiguous [B,H,N,D]");
+  TORCH_CHECK(lse.scalar_type() == at::kFloat && lse.numel() >= (int64_t)B * H * N,
+              "lse must be fp32[B*H*N]");
   const int BH = B * H;
   const int qrows_per_block = kWavesPerBlock * 2 * kQT;
   const int64_t total_blocks = (int64_t)BH * ((N + qrows_per_block - 1) / qrows_per_block);
@@ -605,12 +611,19 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor dout, at::Ten
               double scale, bool causal) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16, "q must be bf16 [B,H,N,D]");
   TORCH_CHECK(q.dim() == 4 && q.size(3) == kAttnD, "bwd supports head dim 64");
+  TORCH_CHECK(k.sizes() == q.sizes() && v.sizes() == q.sizes() && dout.sizes() == q.sizes() &&
+                  o.sizes() == q.sizes(),
+              "self-attention geometry required: k/v/dout/o shapes must equal q");
   TORCH_CHECK(dout.stride(3) == 1 && o.is_contiguous(), "o contiguous; dout last-dim contig");
   TORCH_CHECK(dq.is_contiguous() && dk.is_contiguous() && dv.is_contiguous(),
               "grad outputs must be contiguous");
+  TORCH_CHECK(dq.sizes() == q.sizes() && dk.sizes() == q.sizes() && dv.sizes() == q.sizes(),
+              "grad output shapes must equal q");
   const int B = q.size(0), H = q.size(1), N = q.size(2);
   TORCH_CHECK(N % 64 == 0, "bwd requires N to be a multiple of 64");
   const int BH = B * H;
+  TORCH_CHECK(lse.scalar_type() == at::kFloat && lse.numel() >= (int64_t)BH * N,
+              "lse must be fp32[BH*N]");
   TORCH_CHECK(delta.numel() >= (int64_t)BH * N && delta.scalar_type() == at::kFloat,
               "delta workspace must be fp32[BH*N]");
   auto stream = c10::hip::getCurrentHIPStream();

@@ -30,7 +30,7 @@ void fused_sgd_bf16(at::Tensor param, at::Tensor grad, at::Tensor master,
                     at::Tensor momentum_buf, double lr, double momentum, double weight_decay,
                     double grad_scale, bool use_momentum);
 void l2_norm_and_scale(at::Tensor flat, at::Tensor partials, at::Tensor out, double max_norm,
-                       bool apply);
+                       bool apply, double norm_scale);
 
 // smallcnn.hip
 void conv3x3_relu_pool_fwd(at::Tensor in, at::Tensor w, at::Tensor b, at::Tensor out,
@@ -48,9 +48,10 @@ void layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor mean, at::Tensor rstd
                    at::Tensor dbeta);
 
 // loss.hip
-void ce_fwd(at::Tensor logits, at::Tensor targets, at::Tensor loss, at::Tensor lse);
+void ce_fwd(at::Tensor logits, at::Tensor targets, at::Tensor loss, at::Tensor lse,
+            int64_t ignore_index);
 void ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Tensor scale,
-            at::Tensor dlogits);
+            at::Tensor dlogits, int64_t ignore_index);
 
 // attention.hip (experimental)
 void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor lse,

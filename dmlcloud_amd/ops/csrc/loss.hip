@@ -37,10 +37,15 @@ __device__ __forceinline__ void online_merge(float& m, float& s, float m2, float
   m = mn;
 }
 
-// One block per row; per-row loss and lse.
+// One block per row; per-row loss and lse. Rows whose target equals
+// ignore_index contribute loss 0 (the host wrapper divides by the valid
+// count — F.cross_entropy ignore_index semantics). A target outside
+// [0, V) that is NOT ignore_index writes NaN instead of reading out of
+// bounds: loud, not garbage.
 __global__ void __launch_bounds__(kBlock) ce_fwd_kernel(
     const __hip_bfloat16* __restrict__ logits, const int64_t* __restrict__ targets,
-    float* __restrict__ loss_out, float* __restrict__ lse_out, int64_t R, int64_t V) {
+    float* __restrict__ loss_out, float* __restrict__ lse_out, int64_t R, int64_t V,
+    int64_t ignore_index) {
   __shared__ float lds_m[kBlock / kWave];
   __shared__ float lds_s[kBlock / kWave];
 
@@ -83,28 +88,38 @@ __global__ void __launch_bounds__(kBlock) ce_fwd_kernel(
       for (int w = 1; w < kBlock / kWave; ++w) online_merge(mt, st, lds_m[w], lds_s[w]);
       const float lse = mt + __logf(st);
       lse_out[row] = lse;
-      const float xt = __bfloat162float(((const __hip_bfloat16*)logits)[row * V + targets[row]]);
-      loss_out[row] = lse - xt;
+      const int64_t tgt = targets[row];
+      if (tgt == ignore_index) {
+        loss_out[row] = 0.0f;
+      } else if (tgt < 0 || tgt >= V) {
+        loss_out[row] = __builtin_nanf("");
+      } else {
+        const float xt = __bfloat162float(((const __hip_bfloat16*)logits)[row * V + tgt]);
+        loss_out[row] = lse - xt;
+      }
     }
     __syncthreads();
   }
 }
 
 // dlogits = (exp(x - lse) - onehot) * scale  (scale folds the mean + any
-// upstream grad; elementwise over R x V)
+// upstream grad; elementwise over R x V). Ignored / out-of-range rows
+// write zero gradients.
 __global__ void __launch_bounds__(kBlock) ce_bwd_kernel(
     const __hip_bfloat16* __restrict__ logits, const int64_t* __restrict__ targets,
     const float* __restrict__ lse_in, const float* __restrict__ scale_ptr,
-    __hip_bfloat16* __restrict__ dlogits, int64_t R, int64_t V) {
+    __hip_bfloat16* __restrict__ dlogits, int64_t R, int64_t V, int64_t ignore_index) {
   const float scale = scale_ptr[0];
   const int64_t nvec = V / 8;
   const int64_t stride = (int64_t)gridDim.x * kBlock;
+  const __hip_bfloat16 zero_bf = __float2bfloat16(0.0f);
 
   for (int64_t t = (int64_t)blockIdx.x * kBlock + threadIdx.x; t < R * nvec; t += stride) {
     const int64_t row = t / nvec;
     const int64_t i = t - row * nvec;
     const float lse = lse_in[row];
     const int64_t tgt = targets[row];
+    const bool valid = tgt >= 0 && tgt < V && tgt != ignore_index;
     unsigned short v[8], o[8];
     *(short8*)v = ((const short8*)(logits + row * V))[i];
 #pragma unroll
@@ -112,7 +127,7 @@ __global__ void __launch_bounds__(kBlock) ce_bwd_kernel(
       const int64_t col = i * 8 + k;
       float p = __expf(bf2f_(v[k]) - lse);
       if (col == tgt) p -= 1.0f;
-      const __hip_bfloat16 h = __float2bfloat16(p * scale);
+      const __hip_bfloat16 h = valid ? __float2bfloat16(p * scale) : zero_bf;
       __builtin_memcpy(&o[k], &h, 2);
     }
     ((short8*)(dlogits + row * V))[i] = *(short8*)o;
@@ -122,36 +137,45 @@ __global__ void __launch_bounds__(kBlock) ce_bwd_kernel(
        t += stride) {
     const int64_t row = t / (V - nvec * 8);
     const int64_t col = nvec * 8 + (t - row * (V - nvec * 8));
+    const int64_t tgt = targets[row];
+    const bool valid = tgt >= 0 && tgt < V && tgt != ignore_index;
     float p = __expf(__bfloat162float(((const __hip_bfloat16*)logits)[row * V + col]) -
                      lse_in[row]);
-    if (col == targets[row]) p -= 1.0f;
-    ((__hip_bfloat16*)dlogits)[row * V + col] = __float2bfloat16(p * scale);
+    if (col == tgt) p -= 1.0f;
+    ((__hip_bfloat16*)dlogits)[row * V + col] = valid ? __float2bfloat16(p * scale) : zero_bf;
   }
 }
 
-void ce_fwd(at::Tensor logits, at::Tensor targets, at::Tensor loss, at::Tensor lse) {
+void ce_fwd(at::Tensor logits, at::Tensor targets, at::Tensor loss, at::Tensor lse,
+            int64_t ignore_index) {
   TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kBFloat16 && logits.is_contiguous(),
               "logits must be contiguous bf16");
   TORCH_CHECK(targets.scalar_type() == at::kLong, "targets must be int64");
   const int64_t V = logits.size(-1);
   const int64_t R = logits.numel() / V;
+  TORCH_CHECK(targets.numel() == R, "targets must have one entry per logit row");
+  TORCH_CHECK(loss.scalar_type() == at::kFloat && loss.numel() >= R, "loss must be fp32[R]");
+  TORCH_CHECK(lse.scalar_type() == at::kFloat && lse.numel() >= R, "lse must be fp32[R]");
   auto stream = c10::hip::getCurrentHIPStream();
   const int blocks = (int)std::min<int64_t>(R, kMaxGrid);
   hipLaunchKernelGGL(ce_fwd_kernel, dim3(blocks), dim3(kBlock), 0, stream,
                      (const __hip_bfloat16*)logits.data_ptr(), targets.data_ptr<int64_t>(),
-                     loss.data_ptr<float>(), lse.data_ptr<float>(), R, V);
+                     loss.data_ptr<float>(), lse.data_ptr<float>(), R, V, ignore_index);
 }
 
 void ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Tensor scale,
-            at::Tensor dlogits) {
+            at::Tensor dlogits, int64_t ignore_index) {
   const int64_t V = logits.size(-1);
   const int64_t R = logits.numel() / V;
+  TORCH_CHECK(targets.numel() == R, "targets must have one entry per logit row");
+  TORCH_CHECK(dlogits.sizes() == logits.sizes() && dlogits.scalar_type() == at::kBFloat16,
+              "dlogits must match logits");
   auto stream = c10::hip::getCurrentHIPStream();
   const int blocks = grid_for(R * (V / 8), kBlock);
   hipLaunchKernelGGL(ce_bwd_kernel, dim3(blocks), dim3(kBlock), 0, stream,
                      (const __hip_bfloat16*)logits.data_ptr(), targets.data_ptr<int64_t>(),
                      lse.data_ptr<float>(), scale.data_ptr<float>(),
-                     (__hip_bfloat16*)dlogits.data_ptr(), R, V);
+                     (__hip_bfloat16*)dlogits.data_ptr(), R, V, ignore_index);
 }
 
 } // namespace dmlamd

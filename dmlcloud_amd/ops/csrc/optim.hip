@@ -336,13 +336,18 @@ __global__ void __launch_bounds__(kBlock) scale_bf16_by_device_scalar_kernel(
 
 // Writes norm to out[0] and the clip scale min(1, max_norm/(norm+1e-6)) to
 // out[1] (fp32). max_norm < 0 disables the scale computation (norm only).
+// norm_scale pre-multiplies the computed norm: with a flat gradient buffer
+// that holds the SUM over world ranks, norm_scale = 1/world makes the clip
+// act on the AVERAGED gradient norm — matching torch DDP +
+// clip_grad_norm_ semantics regardless of world size.
 __global__ void __launch_bounds__(kBlock) norm_finalize_kernel(
-    const double* __restrict__ partials, int nblocks, float* __restrict__ out, float max_norm) {
+    const double* __restrict__ partials, int nblocks, float* __restrict__ out, float max_norm,
+    float norm_scale) {
   double local = 0.0;
   for (int i = threadIdx.x; i < nblocks; i += kBlock) local += partials[i];
   double total = block_reduce<double, OP_SUM>(local);
   if (threadIdx.x == 0) {
-    float norm = (float)__builtin_sqrt(total);
+    float norm = (float)__builtin_sqrt(total) * norm_scale;
     out[0] = norm;
     if (max_norm >= 0.0f) {
       float scale = max_norm / (norm + 1e-6f);
@@ -371,9 +376,9 @@ __global__ void __launch_bounds__(kBlock) scale_by_device_scalar_kernel(
 }
 
 // out: fp32[2] {norm, scale}. partials: fp64 workspace (>= grid blocks).
-// flat may be fp32 or bf16.
+// flat may be fp32 or bf16. norm_scale: see norm_finalize_kernel.
 void l2_norm_and_scale(at::Tensor flat, at::Tensor partials, at::Tensor out, double max_norm,
-                       bool apply) {
+                       bool apply, double norm_scale) {
   TORCH_CHECK(flat.is_cuda(), "device tensor required");
   const bool is_bf16 = flat.scalar_type() == at::kBFloat16;
   TORCH_CHECK(is_bf16 || flat.scalar_type() == at::kFloat, "flat must be fp32 or bf16");
@@ -392,7 +397,7 @@ void l2_norm_and_scale(at::Tensor flat, at::Tensor partials, at::Tensor out, dou
   }
   hipLaunchKernelGGL(norm_finalize_kernel, dim3(1), dim3(kBlock), 0, stream,
                      partials.data_ptr<double>(), blocks, out.data_ptr<float>(),
-                     (float)max_norm);
+                     (float)max_norm, (float)norm_scale);
   if (apply) {
     int sblocks = grid_for(n / 4 + 1, kBlock);
     if (is_bf16) {

@@ -1,11 +1,13 @@
-"""EXPERIMENTAL autograd wrapper for the MFMA flash-attention kernels.
+"""Autograd wrapper for the MFMA flash-attention kernels.
 
 `sdpa(q, k, v, causal=True)` matches
 F.scaled_dot_product_attention(..., is_causal=causal) for bf16
-[B, H, N, 64] device tensors; anything else falls back to torch SDPA
-(AOTriton). Kept behind DMLCLOUD_FUSED_ATTN=1 for model use while the
-kernels trail AOTriton (profiles/attention_fwd_experimental.txt);
-numerics are verified in tests/test_gpu.py.
+[B, H, N, 64] self-attention device tensors; anything else (other head
+dims, cross-attention geometry, non-bf16) falls back to torch SDPA
+(AOTriton). The fused path is ON by default whenever usable — set
+DMLCLOUD_DISABLE_FUSED_ATTN=1 to force the AOTriton path (the A/B
+ladder in profiles/ uses this switch). Numerics are verified in
+tests/test_gpu.py.
 """
 
 import math
@@ -44,9 +46,11 @@ class _SdpaFn(torch.autograd.Function):
         return dq, dk, dv, None, None
 
 
-def _usable(q):
+def _usable(q, k, v):
     if os.environ.get('DMLCLOUD_DISABLE_FUSED_ATTN', '0') not in ('0', '', 'false'):
         return False
+    # The kernels assume self-attention geometry: k/v must match q exactly
+    # (cross-attention with a different kv length/head count falls back).
     return (
         q.is_cuda
         and is_available()
@@ -54,6 +58,12 @@ def _usable(q):
         and q.dim() == 4
         and q.shape[-1] == 64
         and q.shape[-2] % 64 == 0
+        and k.shape == q.shape
+        and v.shape == q.shape
+        and k.dtype == q.dtype
+        and v.dtype == q.dtype
+        and k.device == q.device
+        and v.device == q.device
     )
 
 
@@ -68,7 +78,7 @@ def sdpa(q, k, v, causal: bool = True):
     views (e.g. [B,T,H,D] memory viewed as [B,H,T,D]) pass through
     without materialization — only a non-unit last-dim stride forces a
     copy."""
-    if _usable(q):
+    if _usable(q, k, v):
         scale = 1.0 / math.sqrt(q.shape[-1])
         q = q if _lastdim_ok(q) else q.contiguous()
         k = k if _lastdim_ok(k) else k.contiguous()
@@ -78,4 +88,6 @@ def sdpa(q, k, v, causal: bool = True):
 
 
 def fused_attention_enabled() -> bool:
-    return os.environ.get('DMLCLOUD_FUSED_ATTN', '0') not in ('0', '', 'false')
+    """True unless DMLCLOUD_DISABLE_FUSED_ATTN opts out (the fused path
+    is on by default for usable shapes)."""
+    return os.environ.get('DMLCLOUD_DISABLE_FUSED_ATTN', '0') in ('0', '', 'false')

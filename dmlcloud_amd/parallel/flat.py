@@ -22,6 +22,7 @@ This replaces the reference's DDP construction for the hot benchmarks
 still offers torch-DDP semantics for arbitrary models.
 """
 
+from contextlib import contextmanager
 from typing import List, Optional
 
 import torch
@@ -108,6 +109,7 @@ class FlatReplica:
 
         self._buckets = None
         self._pending_works = []
+        self._sync_enabled = True
         if overlap_buckets_mb:
             self._setup_overlap(overlap_buckets_mb)
         if broadcast and dist.is_initialized() and self.world_size > 1:
@@ -157,9 +159,28 @@ class FlatReplica:
         for i, p in enumerate(self.params):
             p.register_post_accumulate_grad_hook(self._make_hook(i))
 
+    @contextmanager
+    def no_sync(self):
+        """Suspend overlap-mode bucket collectives during gradient
+        accumulation (torch DDP ``no_sync`` analog).
+
+        In overlap mode every backward fires async all-reduces from the
+        post-accumulate-grad hooks; two backwards before one step would
+        re-reduce already rank-summed buckets. Wrap all but the LAST
+        micro-batch backward in ``no_sync()`` — local gradients keep
+        accumulating into the flat buffer, and the final (unwrapped)
+        backward all-reduces the accumulated sums once. Single-collective
+        mode needs no guard (sync happens only in grad_sync())."""
+        prev = self._sync_enabled
+        self._sync_enabled = False
+        try:
+            yield
+        finally:
+            self._sync_enabled = prev
+
     def _make_hook(self, index: int):
         def hook(_param):
-            if self.world_size <= 1:
+            if self.world_size <= 1 or not self._sync_enabled:
                 return
             bucket = self._param_bucket[index]
             bucket['done'].add(index)
@@ -213,19 +234,46 @@ class FlatReplica:
         return iter(self.params)
 
 
-class FlatOptimizer:
-    """Base for fused flat-buffer optimizers (single kernel per step)."""
+class FlatOptimizer(torch.optim.Optimizer):
+    """Base for fused flat-buffer optimizers (single kernel per step).
 
-    def __init__(self, replica: FlatReplica):
+    A real ``torch.optim.Optimizer`` subclass: ``param_groups[0]`` holds
+    the live hyperparameters (the fused kernel reads ``lr`` from it each
+    step), so torch LR schedulers work unmodified —
+    ``register_optimizer('opt', FlatAdam(replica), scheduler)`` schedules
+    exactly like the stock-optimizer path. The per-param ``self.state``
+    dict stays empty; optimizer state lives in the flat moment buffers.
+    """
+
+    def __init__(self, replica: FlatReplica, defaults: dict):
         self.replica = replica
-        self.param_groups = [{'params': replica.params, 'lr': None}]  # introspection only
+        super().__init__(replica.params, defaults)
+        if len(self.param_groups) != 1:
+            raise ValueError('FlatOptimizer drives ONE flat buffer = one param group')
+
+    @property
+    def lr(self) -> float:
+        return self.param_groups[0]['lr']
+
+    @lr.setter
+    def lr(self, value: float):
+        self.param_groups[0]['lr'] = value
 
     def zero_grad(self, set_to_none: bool = False):
         self.replica.zero_grad()
 
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
-        """Fused global-norm clip on the flat grads (no host sync)."""
-        return ops.clip_grad_norm_(self.replica.flat_grad, max_norm)
+        """Fused global-norm clip on the flat grads (no host sync).
+
+        After grad_sync() the flat buffer holds the SUM over ranks; the
+        averaging happens later inside the fused optimizer (grad_scale).
+        Passing norm_scale = grad_scale here makes the clip act on the
+        averaged-gradient norm, so the result matches torch DDP +
+        clip_grad_norm_ at any world size (returned norm is the averaged
+        norm too)."""
+        return ops.clip_grad_norm_(
+            self.replica.flat_grad, max_norm, norm_scale=self.replica.grad_scale
+        )
 
     def state_dict(self):
         raise NotImplementedError
@@ -236,19 +284,16 @@ class FlatOptimizer:
 
 class FlatAdam(FlatOptimizer):
     def __init__(self, replica: FlatReplica, lr=1e-3, betas=(0.9, 0.999), eps=1e-8, weight_decay=0.0):
-        super().__init__(replica)
-        self.lr = lr
-        self.beta1, self.beta2 = betas
-        self.eps = eps
-        self.weight_decay = weight_decay
+        super().__init__(replica, dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay))
         n = replica.flat_param.numel()
         dev = replica.flat_param.device
         self.exp_avg = torch.zeros(n, dtype=torch.float32, device=dev)
         self.exp_avg_sq = torch.zeros(n, dtype=torch.float32, device=dev)
         self.step_t = torch.zeros(1, dtype=torch.int32, device=dev)
-        self.param_groups[0]['lr'] = lr
 
-    def step(self):
+    def step(self, closure=None):
+        group = self.param_groups[0]
+        beta1, beta2 = group['betas']
         if self.replica.dtype is torch.bfloat16:
             ops.fused_adam_bf16(
                 self.replica.flat_param,
@@ -257,11 +302,11 @@ class FlatAdam(FlatOptimizer):
                 self.exp_avg,
                 self.exp_avg_sq,
                 self.step_t,
-                self.lr,
-                self.beta1,
-                self.beta2,
-                self.eps,
-                self.weight_decay,
+                group['lr'],
+                beta1,
+                beta2,
+                group['eps'],
+                group['weight_decay'],
                 grad_scale=self.replica.grad_scale,
             )
         else:
@@ -271,20 +316,21 @@ class FlatAdam(FlatOptimizer):
                 self.exp_avg,
                 self.exp_avg_sq,
                 self.step_t,
-                self.lr,
-                self.beta1,
-                self.beta2,
-                self.eps,
-                self.weight_decay,
+                group['lr'],
+                beta1,
+                beta2,
+                group['eps'],
+                group['weight_decay'],
                 grad_scale=self.replica.grad_scale,
             )
 
     def state_dict(self):
+        group = self.param_groups[0]
         return {
-            'lr': self.lr,
-            'betas': (self.beta1, self.beta2),
-            'eps': self.eps,
-            'weight_decay': self.weight_decay,
+            'lr': group['lr'],
+            'betas': group['betas'],
+            'eps': group['eps'],
+            'weight_decay': group['weight_decay'],
             'exp_avg': self.exp_avg,
             'exp_avg_sq': self.exp_avg_sq,
             'step': self.step_t,
@@ -292,10 +338,11 @@ class FlatAdam(FlatOptimizer):
         }
 
     def load_state_dict(self, state):
-        self.lr = state['lr']
-        self.beta1, self.beta2 = state['betas']
-        self.eps = state['eps']
-        self.weight_decay = state['weight_decay']
+        group = self.param_groups[0]
+        group['lr'] = state['lr']
+        group['betas'] = tuple(state['betas'])
+        group['eps'] = state['eps']
+        group['weight_decay'] = state['weight_decay']
         self.exp_avg.copy_(state['exp_avg'].to(self.exp_avg.device))
         self.exp_avg_sq.copy_(state['exp_avg_sq'].to(self.exp_avg_sq.device))
         self.step_t.copy_(state['step'].to(self.step_t.device))
@@ -305,28 +352,25 @@ class FlatAdam(FlatOptimizer):
 
 class FlatSGD(FlatOptimizer):
     def __init__(self, replica: FlatReplica, lr=1e-2, momentum=0.0, weight_decay=0.0):
-        super().__init__(replica)
-        self.lr = lr
-        self.momentum = momentum
-        self.weight_decay = weight_decay
+        super().__init__(replica, dict(lr=lr, momentum=momentum, weight_decay=weight_decay))
         self.momentum_buf: Optional[torch.Tensor] = None
         if momentum != 0:
             # momentum accumulates in fp32 regardless of the param dtype
             self.momentum_buf = torch.zeros(
                 replica.flat_param.numel(), dtype=torch.float32, device=replica.flat_param.device
             )
-        self.param_groups[0]['lr'] = lr
 
-    def step(self):
+    def step(self, closure=None):
+        group = self.param_groups[0]
         if self.replica.dtype is torch.bfloat16:
             ops.fused_sgd_bf16(
                 self.replica.flat_param,
                 self.replica.flat_grad,
                 self.replica.flat_master,
                 self.momentum_buf,
-                self.lr,
-                self.momentum,
-                self.weight_decay,
+                group['lr'],
+                group['momentum'],
+                group['weight_decay'],
                 grad_scale=self.replica.grad_scale,
             )
         else:
@@ -334,25 +378,27 @@ class FlatSGD(FlatOptimizer):
                 self.replica.flat_param,
                 self.replica.flat_grad,
                 self.momentum_buf,
-                self.lr,
-                self.momentum,
-                self.weight_decay,
+                group['lr'],
+                group['momentum'],
+                group['weight_decay'],
                 grad_scale=self.replica.grad_scale,
             )
 
     def state_dict(self):
+        group = self.param_groups[0]
         return {
-            'lr': self.lr,
-            'momentum': self.momentum,
-            'weight_decay': self.weight_decay,
+            'lr': group['lr'],
+            'momentum': group['momentum'],
+            'weight_decay': group['weight_decay'],
             'momentum_buf': self.momentum_buf,
             'master': self.replica.flat_master,
         }
 
     def load_state_dict(self, state):
-        self.lr = state['lr']
-        self.momentum = state['momentum']
-        self.weight_decay = state['weight_decay']
+        group = self.param_groups[0]
+        group['lr'] = state['lr']
+        group['momentum'] = state['momentum']
+        group['weight_decay'] = state['weight_decay']
         if state['momentum_buf'] is not None:
             if self.momentum_buf is None:
                 self.momentum_buf = torch.zeros(

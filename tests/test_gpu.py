@@ -215,6 +215,25 @@ class TestOptimKernels:
         assert norm.item() == pytest.approx(norm_ref.item(), rel=1e-4)
         torch.testing.assert_close(g.cpu(), p.grad, rtol=1e-4, atol=1e-7)
 
+    def test_clip_norm_scale(self):
+        """norm_scale emulates the 1/world averaging: clipping a W-summed
+        gradient with norm_scale=1/W == clipping the averaged gradient."""
+        _assert_native_loaded()
+        torch.manual_seed(3)
+        world = 4
+        g_avg = torch.randn(100_000, device=DEV)
+        g_sum = g_avg * world
+
+        norm = ops.clip_grad_norm_(g_sum, 0.5, norm_scale=1.0 / world)
+        # reported norm is the averaged norm
+        assert norm.item() == pytest.approx(g_avg.norm().item(), rel=1e-4)
+
+        p = torch.nn.Parameter(torch.zeros_like(g_avg))
+        p.grad = g_avg.clone()
+        torch.nn.utils.clip_grad_norm_([p], 0.5)
+        # after the later 1/W averaging, the clipped sum equals torch's clipped average
+        torch.testing.assert_close(g_sum / world, p.grad, rtol=1e-4, atol=1e-7)
+
 
 class TestCopyKernel:
     def test_pack_unpack_roundtrip(self):
@@ -442,6 +461,38 @@ class TestFusedCE:
         F.cross_entropy(l2, targets).backward()
         torch.testing.assert_close(logits.grad.float(), l2.grad, rtol=5e-2, atol=1e-5)
 
+    def test_ignore_index_matches_torch(self):
+        """ignore_index rows contribute no loss/grad; mean divides by the
+        valid count (padded-token safety)."""
+        import torch.nn.functional as F
+
+        from dmlcloud_amd.ops.fused_loss import cross_entropy
+
+        torch.manual_seed(2)
+        rows, v = 64, 1000
+        logits = (torch.randn(rows, v, device=DEV) * 2).to(torch.bfloat16).requires_grad_(True)
+        targets = torch.randint(0, v, (rows,), device=DEV)
+        targets[::3] = -100  # every third row padded
+        loss = cross_entropy(logits, targets)
+        loss.backward()
+
+        l2 = logits.detach().float().requires_grad_(True)
+        ref = F.cross_entropy(l2, targets)
+        ref.backward()
+        assert loss.item() == pytest.approx(ref.item(), rel=2e-3)
+        torch.testing.assert_close(logits.grad.float(), l2.grad, rtol=5e-2, atol=1e-5)
+        # ignored rows have exactly zero gradient
+        assert logits.grad[::3].abs().max().item() == 0.0
+
+    def test_out_of_range_target_is_loud(self):
+        """A target >= V (not ignore_index) produces NaN loss, not an OOB read."""
+        from dmlcloud_amd.ops.fused_loss import cross_entropy
+
+        logits = torch.randn(4, 100, device=DEV).to(torch.bfloat16)
+        targets = torch.tensor([0, 1, 100, 2], device=DEV)  # 100 out of range
+        loss = cross_entropy(logits, targets)
+        assert torch.isnan(loss).item()
+
     def test_gpt2_tiny_step_matches_eager(self):
         """Full fused-GPT2 (LN + CE) loss matches a plain-torch computation."""
         from dmlcloud_amd.models import gpt2_tiny
@@ -539,6 +590,20 @@ class TestAttnFwdExperimental:
         torch.testing.assert_close(q.grad, q2.grad)
         torch.testing.assert_close(k.grad, k2.grad)
         torch.testing.assert_close(v.grad, v2.grad)
+
+    def test_cross_attention_falls_back(self):
+        """Different kv length (cross-attention geometry) must take the
+        torch SDPA fallback, not feed the self-attention kernels
+        (OOB read)."""
+        from dmlcloud_amd.ops.fused_attn import sdpa
+
+        torch.manual_seed(7)
+        q = (torch.randn(2, 4, 128, 64, device=DEV) * 0.5).to(torch.bfloat16)
+        k = (torch.randn(2, 4, 256, 64, device=DEV) * 0.5).to(torch.bfloat16)
+        v = (torch.randn(2, 4, 256, 64, device=DEV) * 0.5).to(torch.bfloat16)
+        out = sdpa(q, k, v, causal=False)
+        ref = torch.nn.functional.scaled_dot_product_attention(q.float(), k.float(), v.float(), is_causal=False)
+        torch.testing.assert_close(out.float(), ref, rtol=3e-2, atol=3e-2)
 
     def test_lse_matches_blueprint(self):
         import math

@@ -172,5 +172,73 @@ class TestFlatBf16Cpu:
         assert pipeline.tracker['train/loss'][0] is not None
 
 
+class TestFlatLRScheduling:
+    """FlatOptimizer is a real torch.optim.Optimizer: torch LR schedulers
+    drive the fused step and misc/lr_* tracks (VERDICT r1 next-round #5)."""
+
+    def test_flat_optimizer_is_torch_optimizer(self, torch_distributed):
+        from dmlcloud_amd.parallel import FlatAdam, FlatReplica, FlatSGD
+
+        replica = FlatReplica(torch.nn.Linear(4, 2))
+        assert isinstance(FlatSGD(replica, lr=0.1), torch.optim.Optimizer)
+        replica2 = FlatReplica(torch.nn.Linear(4, 2))
+        opt = FlatAdam(replica2, lr=1e-3)
+        assert isinstance(opt, torch.optim.Optimizer)
+        assert opt.param_groups[0]['lr'] == pytest.approx(1e-3)
+        assert opt.lr == pytest.approx(1e-3)
+
+    def test_steplr_drives_fused_step(self, torch_distributed):
+        from dmlcloud_amd.parallel import FlatReplica, FlatSGD
+
+        torch.manual_seed(0)
+        replica = FlatReplica(torch.nn.Linear(4, 2))
+        opt = FlatSGD(replica, lr=0.1)
+        sched = torch.optim.lr_scheduler.StepLR(opt, step_size=1, gamma=0.5)
+
+        lrs = []
+        for _ in range(3):
+            replica.zero_grad()
+            replica(torch.ones(2, 4)).sum().backward()
+            opt.step()
+            lrs.append(opt.param_groups[0]['lr'])
+            sched.step()
+        assert lrs == pytest.approx([0.1, 0.05, 0.025])
+
+    def test_pipeline_tracks_scheduled_lr(self, torch_distributed):
+        from dmlcloud_amd.parallel import FlatSGD
+
+        class SchedStage(Stage_):
+            def pre_stage(self):
+                torch.manual_seed(0)
+                model = torch.nn.Linear(10, 10)
+                self.pipeline.register_model('m', model, ddp_impl='flat')
+                replica = self.pipeline.models['m']
+                opt = FlatSGD(replica, lr=0.1)
+                sched = torch.optim.lr_scheduler.StepLR(opt, step_size=1, gamma=0.1)
+                self.pipeline.register_optimizer('opt', opt, sched)
+                self.pipeline.register_dataset('train', torch.utils.data.DataLoader(DS(), batch_size=4))
+                self.pipeline.register_dataset('val', torch.utils.data.DataLoader(DS(), batch_size=4))
+                self.loss = torch.nn.CrossEntropyLoss()
+
+        pipeline = TrainingPipeline()
+        pipeline.append_stage(SchedStage(), max_epochs=3)
+        pipeline.run()
+        lr_history = pipeline.tracker['misc/lr_opt']
+        assert lr_history[0] == pytest.approx(0.1)
+        assert lr_history[1] == pytest.approx(0.01)
+        assert lr_history[2] == pytest.approx(0.001)
+
+
+class TestFusedCEIgnoreIndex:
+    def test_cpu_fallback_matches_torch(self):
+        from dmlcloud_amd.ops.fused_loss import cross_entropy
+
+        torch.manual_seed(0)
+        logits = torch.randn(6, 11)
+        targets = torch.tensor([0, 3, -100, 5, -100, 10])
+        expected = torch.nn.functional.cross_entropy(logits, targets)
+        torch.testing.assert_close(cross_entropy(logits, targets), expected)
+
+
 if __name__ == '__main__':
     sys.exit(pytest.main([__file__]))
