@@ -6,6 +6,12 @@ xarray dataset sharding (optional dependency, duck-typed), DataLoader
 worker-id folding (worker rank = rank * num_workers + worker_id),
 prefetch/batch wrappers, and batch interleaving.
 
+Design: sharding is expressed as two tiny primitives — a seeded
+permutation and a strided split — composed by everything else, so the
+exact element orders (the behavioral contract the reference's test
+matrix pins down) live in one place. The iterable-dataset wrappers share
+one `_shard_context()` helper for worker-id folding.
+
 MI355X-native difference: on device tensors, interleave_batches /
 interleave_dict_batches execute as ONE descriptor-table gather kernel
 (ops/csrc/copy.hip) instead of the reference's N^2 Python slice-copy loop
@@ -13,8 +19,9 @@ interleave_dict_batches execute as ONE descriptor-table gather kernel
 side stream.
 """
 
-from concurrent.futures import ThreadPoolExecutor
-from typing import Iterable, List, Optional, Sequence
+import queue
+import threading
+from typing import Iterable, Iterator, List, Optional, Sequence, Tuple
 
 import numpy as np
 import torch
@@ -23,13 +30,47 @@ from torch.utils.data import IterableDataset, get_worker_info
 
 from . import ops
 
-try:
-    import xarray as xr
+__all__ = [
+    'shard_indices',
+    'chunk_and_shard_indices',
+    'shard_sequence',
+    'sharded_xr_dataset',
+    'ShardedSequenceDataset',
+    'ShardedXrDataset',
+    'DownstreamDataset',
+    'PrefetchDataset',
+    'BatchDataset',
+    'interleave_batches',
+    'interleave_dict_batches',
+]
 
-    _HAS_XARRAY = True
-except ImportError:
-    xr = None
-    _HAS_XARRAY = False
+
+# ------------------------------------------------------------- primitives
+
+
+def _element_order(count: int, shuffle: bool, seed: int) -> np.ndarray:
+    """0..count-1, optionally permuted by a seeded MT19937 generator.
+
+    MT19937 (not the numpy default PCG64) is deliberate: the permutation
+    IS the cross-rank data-assignment contract, and every rank must
+    derive the identical order from the same seed on any platform.
+    """
+    order = np.arange(count)
+    if shuffle:
+        np.random.Generator(np.random.MT19937(seed)).shuffle(order)
+    return order
+
+
+def _strided_split(order: np.ndarray, rank: int, world_size: int, even_shards: bool) -> np.ndarray:
+    """Take every world_size-th element starting at `rank`.
+
+    even_shards trims the tail first so every rank gets exactly
+    len(order) // world_size elements (remainder dropped).
+    """
+    if even_shards:
+        usable = len(order) - len(order) % world_size
+        order = order[:usable]
+    return order[rank::world_size]
 
 
 def shard_indices(
@@ -40,17 +81,9 @@ def shard_indices(
     even_shards: bool = True,
     seed: int = 0,
 ) -> List[int]:
-    """Strided [rank::world_size] sharding.
-
-    even_shards: every worker receives the same number of elements; the
-    trailing remainder is dropped.
-    """
-    indices = np.arange(num_elements)
-    if shuffle:
-        np.random.Generator(np.random.MT19937(seed)).shuffle(indices)
-    if even_shards:
-        indices = indices[: num_elements - num_elements % world_size]
-    return indices[rank::world_size].tolist()
+    """This rank's element indices under strided [rank::world_size] sharding."""
+    order = _element_order(num_elements, shuffle, seed)
+    return _strided_split(order, rank, world_size, even_shards).tolist()
 
 
 def chunk_and_shard_indices(
@@ -63,21 +96,22 @@ def chunk_and_shard_indices(
     equal_chunks: bool = True,
     shuffle: bool = False,
     seed: int = 0,
-):
+) -> List[Tuple[int, int]]:
     """Shard half-open (start, start+chunk_size+overlap) windows across
-    ranks — e.g. temporal windows of weather data."""
+    ranks — e.g. temporal windows of weather data.
+
+    equal_chunks=True keeps only full-size chunks (floor division);
+    False also emits a final short chunk covering the tail.
+    """
     if equal_chunks:
         num_chunks = num_elements // chunk_size
     else:
-        num_chunks = (num_elements + chunk_size - 1) // chunk_size
-
-    chunk_indices = shard_indices(num_chunks, rank, world_size, shuffle=shuffle, even_shards=even_shards, seed=seed)
-    chunks = []
-    for chunk_idx in chunk_indices:
-        start = chunk_idx * chunk_size
-        end = start + chunk_size + chunk_overlap
-        chunks.append((start, end))
-    return chunks
+        num_chunks = -(-num_elements // chunk_size)  # ceil
+    my_chunks = shard_indices(
+        num_chunks, rank, world_size, shuffle=shuffle, even_shards=even_shards, seed=seed
+    )
+    window = chunk_size + chunk_overlap
+    return [(c * chunk_size, c * chunk_size + window) for c in my_chunks]
 
 
 def shard_sequence(
@@ -87,9 +121,12 @@ def shard_sequence(
     shuffle: bool = False,
     even_shards: bool = True,
     seed: int = 0,
-):
-    indices = shard_indices(len(sequence), rank, world_size, shuffle=shuffle, even_shards=even_shards, seed=seed)
-    return [sequence[i] for i in indices]
+) -> list:
+    """Materialize this rank's shard of an arbitrary sequence."""
+    picks = shard_indices(
+        len(sequence), rank, world_size, shuffle=shuffle, even_shards=even_shards, seed=seed
+    )
+    return [sequence[i] for i in picks]
 
 
 def sharded_xr_dataset(
@@ -106,7 +143,7 @@ def sharded_xr_dataset(
     process_group=None,
     load: bool = False,
     load_kwargs: Optional[dict] = None,
-) -> Iterable:
+) -> Iterator:
     """Yield rank-sharded chunks of an xarray Dataset/DataArray along `dim`.
 
     Duck-typed: any object with ``len(ds[dim])`` and ``.isel({dim: slice})``
@@ -117,9 +154,8 @@ def sharded_xr_dataset(
     if world_size is None:
         world_size = dist.get_world_size(process_group)
 
-    num_elements = len(ds[dim])
-    chunks = chunk_and_shard_indices(
-        num_elements,
+    windows = chunk_and_shard_indices(
+        len(ds[dim]),
         chunk_size,
         rank,
         world_size,
@@ -129,23 +165,56 @@ def sharded_xr_dataset(
         shuffle=shuffle,
         seed=seed,
     )
-    for start, end in chunks:
-        chunk = ds.isel({dim: slice(start, end)})
+    for start, stop in windows:
+        piece = ds.isel({dim: slice(start, stop)})
         if load:
-            chunk.load(**(load_kwargs or {}))
-        yield chunk
+            piece.load(**(load_kwargs or {}))
+        yield piece
 
 
-def _effective_rank(rank: int, world_size: int):
-    """Fold the DataLoader worker id into the rank so loader workers shard
-    disjointly: worker rank = rank * num_workers + worker_id."""
-    worker_info = get_worker_info()
-    if worker_info is None:
+# ----------------------------------------------------- iterable datasets
+
+
+def _shard_context(rank: int, world_size: int) -> Tuple[int, int]:
+    """(effective_rank, effective_world) after DataLoader worker folding.
+
+    Inside a DataLoader worker process, each of the `num_workers` workers
+    of each rank must own a disjoint slice, so the worker id extends the
+    rank: effective rank = rank * num_workers + worker_id, effective
+    world = world_size * num_workers. Outside a worker this is identity.
+    """
+    info = get_worker_info()
+    if info is None:
         return rank, world_size
-    return rank * worker_info.num_workers + worker_info.id, world_size * worker_info.num_workers
+    return rank * info.num_workers + info.id, world_size * info.num_workers
 
 
-class ShardedSequenceDataset(IterableDataset):
+def _resolve_rank_world(rank, world_size, process_group=None) -> Tuple[int, int]:
+    if rank is None:
+        rank = dist.get_rank(process_group)
+    if world_size is None:
+        world_size = dist.get_world_size(process_group)
+    return rank, world_size
+
+
+class _EpochSeeded(IterableDataset):
+    """Shared bits of the sharded iterable datasets: the epoch counter
+    advances the shuffle seed so each epoch draws a fresh permutation."""
+
+    def __init__(self):
+        self._epoch = 0
+
+    def set_epoch(self, epoch: int):
+        self._epoch = epoch
+
+    @property
+    def epoch(self) -> int:
+        return self._epoch
+
+
+class ShardedSequenceDataset(_EpochSeeded):
+    """Iterable view of this rank's (and loader-worker's) shard of a sequence."""
+
     def __init__(
         self,
         sequence: Sequence,
@@ -155,37 +224,38 @@ class ShardedSequenceDataset(IterableDataset):
         rank: Optional[int] = None,
         world_size: Optional[int] = None,
     ):
+        super().__init__()
         self.sequence = sequence
         self.shuffle = shuffle
         self.even_shards = even_shards
         self.seed = seed
-        self.rank = rank if rank is not None else dist.get_rank()
-        self.world_size = world_size if world_size is not None else dist.get_world_size()
-        self.epoch = 0
-
-    def set_epoch(self, epoch: int):
-        self.epoch = epoch
+        self.rank, self.world_size = _resolve_rank_world(rank, world_size)
 
     def __len__(self):
-        n = len(self.sequence)
+        total = len(self.sequence)
         if self.even_shards:
-            return (n - n % self.world_size) // self.world_size
-        return (n - self.rank + self.world_size - 1) // self.world_size
+            return total // self.world_size
+        # uneven: ranks below the remainder get one extra element
+        base, extra = divmod(total, self.world_size)
+        return base + (1 if self.rank < extra else 0)
 
     def __iter__(self):
-        rank, world_size = _effective_rank(self.rank, self.world_size)
-        shards = shard_sequence(
-            self.sequence,
-            rank,
-            world_size,
-            shuffle=self.shuffle,
-            even_shards=self.even_shards,
-            seed=self.seed + self.epoch,
+        rank, world = _shard_context(self.rank, self.world_size)
+        return iter(
+            shard_sequence(
+                self.sequence,
+                rank,
+                world,
+                shuffle=self.shuffle,
+                even_shards=self.even_shards,
+                seed=self.seed + self.epoch,
+            )
         )
-        return iter(shards)
 
 
-class ShardedXrDataset(IterableDataset):
+class ShardedXrDataset(_EpochSeeded):
+    """Iterable view of this rank's chunk windows of an xarray dataset."""
+
     def __init__(
         self,
         ds,
@@ -202,6 +272,7 @@ class ShardedXrDataset(IterableDataset):
         load: bool = False,
         load_kwargs: Optional[dict] = None,
     ):
+        super().__init__()
         self.ds = ds
         self.dim = dim
         self.chunk_size = chunk_size
@@ -212,16 +283,10 @@ class ShardedXrDataset(IterableDataset):
         self.seed = seed
         self.load = load
         self.load_kwargs = load_kwargs
-
-        self.rank = rank if rank is not None else dist.get_rank(process_group)
-        self.world_size = world_size if world_size is not None else dist.get_world_size(process_group)
-        self._num_iters = 0
-
-    def set_epoch(self, epoch: int):
-        self._num_iters = epoch
+        self.rank, self.world_size = _resolve_rank_world(rank, world_size, process_group)
 
     def __iter__(self):
-        rank, world_size = _effective_rank(self.rank, self.world_size)
+        rank, world = _shard_context(self.rank, self.world_size)
         return sharded_xr_dataset(
             self.ds,
             self.dim,
@@ -230,15 +295,17 @@ class ShardedXrDataset(IterableDataset):
             even_shards=self.even_shards,
             equal_chunks=self.equal_chunks,
             shuffle=self.shuffle,
-            seed=self.seed + self._num_iters,
+            seed=self.seed + self.epoch,
             rank=rank,
-            world_size=world_size,
+            world_size=world,
             load=self.load,
             load_kwargs=self.load_kwargs,
         )
 
 
 class DownstreamDataset(IterableDataset):
+    """Wrapper base: forwards set_epoch/len to the wrapped iterable."""
+
     def __init__(self, source_ds: Iterable):
         self.source_ds = source_ds
 
@@ -251,47 +318,66 @@ class DownstreamDataset(IterableDataset):
 
 
 class PrefetchDataset(DownstreamDataset):
-    """Background-thread lookahead of `num_elements` items."""
+    """Decouple producer latency from the training loop: a daemon thread
+    stays `num_elements` items ahead in a bounded queue."""
+
+    _DONE = object()
 
     def __init__(self, source_ds: Iterable, num_elements: int):
         super().__init__(source_ds)
         self.num_elements = num_elements
 
     def __iter__(self):
-        pool = ThreadPoolExecutor(max_workers=1)
-        iter_ = iter(self.source_ds)
-        with pool:
-            futures = [pool.submit(next, iter_) for _ in range(self.num_elements)]
-            while True:
-                future = futures.pop(0)
-                try:
-                    element = future.result()
-                except StopIteration:
-                    return
-                futures += [pool.submit(next, iter_)]
-                yield element
+        buffer: queue.Queue = queue.Queue(maxsize=self.num_elements)
+        error = []
+
+        def producer():
+            try:
+                for item in self.source_ds:
+                    buffer.put(item)
+            except BaseException as e:  # surfaced in the consumer
+                error.append(e)
+            finally:
+                buffer.put(self._DONE)
+
+        worker = threading.Thread(target=producer, daemon=True)
+        worker.start()
+        while True:
+            item = buffer.get()
+            if item is self._DONE:
+                worker.join()
+                if error:
+                    raise error[0]
+                return
+            yield item
 
 
 class BatchDataset(DownstreamDataset):
+    """Group consecutive elements into lists of `batch_size`."""
+
     def __init__(self, source_ds: Iterable, batch_size: int, drop_remainder: bool = False):
         super().__init__(source_ds)
         self.batch_size = batch_size
         self.drop_remainder = drop_remainder
 
     def __len__(self):
-        if self.drop_remainder:
-            return len(self.source_ds) // self.batch_size
-        return (len(self.source_ds) + self.batch_size - 1) // self.batch_size
+        full, rest = divmod(len(self.source_ds), self.batch_size)
+        if rest and not self.drop_remainder:
+            return full + 1
+        return full
 
     def __iter__(self):
-        batch = []
+        pending = []
         for element in self.source_ds:
-            batch.append(element)
-            if len(batch) == self.batch_size:
-                yield batch
-                batch = []
-        if batch and not self.drop_remainder:
-            yield batch
+            pending.append(element)
+            if len(pending) == self.batch_size:
+                yield pending
+                pending = []
+        if pending and not self.drop_remainder:
+            yield pending
+
+
+# ----------------------------------------------------------- interleaving
 
 
 def _interleave_group(batches: List[torch.Tensor], memory: torch.Tensor, slice_size: int):

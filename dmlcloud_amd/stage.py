@@ -5,18 +5,24 @@ dmlcloud/stage.py:18-341): same hook order (pre_stage -> [pre_epoch ->
 run_epoch -> post_epoch]* -> post_stage), same metric namespacing
 (train/ val/ misc/), same table semantics and stop_stage behavior.
 
-Differences by design:
-- progress table is the dependency-free utils/table.py implementation,
-- per-step metric tracking feeds device-resident accumulators (no D2H
-  per batch - see metrics.py),
+Internal architecture (this implementation's own):
+- table columns are normalized once into ``_ColumnSpec`` records; the
+  live table rendering and the per-epoch refresh both work off that list,
+- epoch/stage wall-time flows through one ``_Clock`` helper,
+- the per-batch hot path is ``TrainValStage.train_batch`` — a single
+  method the benchmark driver can call directly (and capture into a
+  hipGraph via parallel/graphs.py),
+- metric tracking feeds device-resident accumulators (no D2H per batch,
+  see metrics.py),
 - clip_gradients() computes a fused global-norm clip on flat optimizers
-  (single gfx950 kernel chain, no host sync) and falls back to
-  torch.nn.utils for stock optimizers.
+  (single gfx950 kernel chain, no host sync, correct at any world size)
+  and falls back to torch.nn.utils for stock optimizers.
 """
 
 import sys
 import time
-from datetime import datetime
+from dataclasses import dataclass
+from datetime import datetime, timedelta
 from typing import Any, Dict, List, Optional, Union
 
 import torch
@@ -31,6 +37,44 @@ from .utils.tracing import roctx_range
 __all__ = ['Stage', 'TrainValStage']
 
 
+@dataclass
+class _ColumnSpec:
+    """One progress-table column: display name + the tracked metric that
+    feeds it (None = code updates the cell directly, e.g. ETA)."""
+
+    title: str
+    metric: Optional[str]
+    extra: Dict[str, Any]
+
+    @classmethod
+    def parse(cls, raw: Union[str, Dict[str, Any]]) -> '_ColumnSpec':
+        if isinstance(raw, str):
+            return cls(title=raw, metric=raw, extra={})
+        if isinstance(raw, dict):
+            missing = {'name', 'metric'} - raw.keys()
+            if missing:
+                raise ValueError(f'Column dict must contain a "{missing.pop()}" key')
+            extra = {k: v for k, v in raw.items() if k not in ('name', 'metric')}
+            return cls(title=raw['name'], metric=raw['metric'], extra=extra)
+        raise ValueError(f'Invalid column: {raw}. Must be a string or a dict.')
+
+
+class _Clock:
+    """Wall-time bookkeeping for a stage run."""
+
+    def __init__(self):
+        self.stage_started: Optional[datetime] = None
+        self.stage_stopped: Optional[datetime] = None
+        self.epoch_started: Optional[datetime] = None
+        self.epoch_stopped: Optional[datetime] = None
+
+    def stage_elapsed(self) -> timedelta:
+        return datetime.now() - self.stage_started
+
+    def epoch_seconds(self) -> float:
+        return (self.epoch_stopped - self.epoch_started).total_seconds()
+
+
 class Stage:
     """Hook points: pre_stage, post_stage, pre_epoch, post_epoch."""
 
@@ -39,16 +83,16 @@ class Stage:
         self.max_epochs = None
         self.name = None
 
-        self.start_time = None
-        self.stop_time = None
-        self.epoch_start_time = None
-        self.epoch_stop_time = None
         self.current_epoch = 1
-        self._stop_requested = False
-
         self.metric_prefix = None
         self.table = None
         self.barrier_timeout = None
+
+        self._clock = _Clock()
+        self._stop_requested = False
+        self._columns: List[_ColumnSpec] = []
+
+    # ------------------------------------------------- pipeline plumbing
 
     @property
     def tracker(self) -> MetricTracker:
@@ -66,6 +110,25 @@ class Stage:
     def config(self):
         return self.pipeline.config
 
+    # timing attributes kept as properties for API compatibility
+    @property
+    def start_time(self):
+        return self._clock.stage_started
+
+    @property
+    def stop_time(self):
+        return self._clock.stage_stopped
+
+    @property
+    def epoch_start_time(self):
+        return self._clock.epoch_started
+
+    @property
+    def epoch_stop_time(self):
+        return self._clock.epoch_stopped
+
+    # ------------------------------------------------------- user surface
+
     def track_reduce(
         self,
         name: str,
@@ -76,16 +139,18 @@ class Stage:
         reduce_globally: bool = True,
         prefixed: bool = True,
     ):
-        if prefixed and self.metric_prefix:
-            name = f'{self.metric_prefix}/{name}'
-        self.pipeline.track_reduce(name, value, step, reduction, dim, reduce_globally)
+        self.pipeline.track_reduce(self._qualify(name, prefixed), value, step, reduction, dim, reduce_globally)
 
     def track(self, name: str, value, step: Optional[int] = None, prefixed: bool = True):
+        self.pipeline.track(self._qualify(name, prefixed), value, step)
+
+    def _qualify(self, name: str, prefixed: bool) -> str:
         if prefixed and self.metric_prefix:
-            name = f'{self.metric_prefix}/{name}'
-        self.pipeline.track(name, value, step)
+            return f'{self.metric_prefix}/{name}'
+        return name
 
     def stop_stage(self):
+        """Request the epoch loop to end after the current epoch."""
         self._stop_requested = True
 
     def pre_stage(self):
@@ -115,95 +180,72 @@ class Stage:
             columns.append({'name': 'ETA', 'metric': None})
         return columns
 
+    # ----------------------------------------------------------- lifecycle
+
     def run(self):
         """Run until max_epochs or stop_stage()."""
-        self._pre_stage()
-        while self.max_epochs is None or self.current_epoch <= self.max_epochs:
-            self._pre_epoch()
+        self._enter_stage()
+        while not self._epochs_done():
+            self._enter_epoch()
             with roctx_range(f'epoch_{self.current_epoch}'):
                 self.run_epoch()
-            self._post_epoch()
-            if self._stop_requested:
-                break
-        self._post_stage()
+            self._leave_epoch()
+        self._leave_stage()
 
-    def _pre_stage(self):
-        self.start_time = datetime.now()
+    def _epochs_done(self) -> bool:
+        if self._stop_requested:
+            return True
+        return self.max_epochs is not None and self.current_epoch > self.max_epochs
+
+    def _enter_stage(self):
+        self._clock.stage_started = datetime.now()
         self.table = ProgressTable(file=sys.stdout if is_root() else DevNullIO())
-        self._setup_table()
+        self._columns = [_ColumnSpec.parse(c) for c in self.table_columns()]
+        for spec in self._columns:
+            self.table.add_column(spec.title, **spec.extra)
         if len(self.pipeline.stages) > 1:
             self.logger.info(f'\n========== STAGE: {self.name} ==========')
         self.pre_stage()
         flush_log_handlers(self.logger)
         self.pipeline.barrier(self.barrier_timeout)
 
-    def _post_stage(self):
+    def _leave_stage(self):
         self.table.close()
         self.post_stage()
         self.pipeline.barrier(self.barrier_timeout)
-        self.stop_time = datetime.now()
+        self._clock.stage_stopped = datetime.now()
         if len(self.pipeline.stages) > 1:
-            self.logger.info(f'Finished stage in {self.stop_time - self.start_time}')
+            elapsed = self._clock.stage_stopped - self._clock.stage_started
+            self.logger.info(f'Finished stage in {elapsed}')
 
-    def _pre_epoch(self):
-        self.epoch_start_time = datetime.now()
+    def _enter_epoch(self):
+        self._clock.epoch_started = datetime.now()
         self.table['Epoch'] = self.current_epoch
         self.pre_epoch()
         self.pipeline._pre_epoch()
 
-    def _post_epoch(self):
-        self.epoch_stop_time = datetime.now()
-        self._reduce_metrics()
+    def _leave_epoch(self):
+        self._clock.epoch_stopped = datetime.now()
+        self.track(name='misc/epoch', value=self.current_epoch, prefixed=False)
+        self.track(name='misc/epoch_time', value=self._clock.epoch_seconds(), prefixed=False)
+        self.tracker.next_epoch()
         self.post_epoch()
         self.pipeline._post_epoch()
-        self._update_table()
+        self._refresh_table()
         self.current_epoch += 1
 
-    def _reduce_metrics(self):
-        self.track(name='misc/epoch', value=self.current_epoch, prefixed=False)
-        self.track(
-            name='misc/epoch_time',
-            value=(self.epoch_stop_time - self.epoch_start_time).total_seconds(),
-            prefixed=False,
-        )
-        self.tracker.next_epoch()
-
-    def _setup_table(self):
-        for column_dct in self._metrics():
-            display_name = column_dct.pop('name')
-            column_dct.pop('metric')
-            self.table.add_column(display_name, **column_dct)
-
-    def _update_table(self):
+    def _refresh_table(self):
+        per_epoch = self._clock.stage_elapsed() / self.current_epoch
         self.table.update('Epoch', self.current_epoch)
-        self.table.update('Time/Epoch', (datetime.now() - self.start_time) / self.current_epoch)
+        self.table.update('Time/Epoch', per_epoch)
         if self.max_epochs is not None:
-            self.table.update(
-                'ETA',
-                (datetime.now() - self.start_time) / self.current_epoch * (self.max_epochs - self.current_epoch),
-            )
-        for column_dct in self._metrics():
-            display_name = column_dct['name']
-            metric_name = column_dct['metric']
-            if metric_name is not None:
-                history = self.tracker[metric_name]
-                self.table.update(display_name, history[-1] if history else None)
+            self.table.update('ETA', per_epoch * (self.max_epochs - self.current_epoch))
+        for spec in self._columns:
+            if spec.metric is None:
+                continue
+            history = self.tracker[spec.metric]
+            self.table.update(spec.title, history[-1] if history else None)
         self.table.next_row()
-
-    def _metrics(self):
-        metrics = []
-        for column in self.table_columns():
-            if isinstance(column, str):
-                metrics.append({'name': column, 'metric': column})
-            elif isinstance(column, dict):
-                if 'name' not in column:
-                    raise ValueError('Column dict must contain a "name" key')
-                if 'metric' not in column:
-                    raise ValueError('Column dict must contain a "metric" key')
-                metrics.append(dict(column))
-            else:
-                raise ValueError(f'Invalid column: {column}. Must be a string or a dict.')
-        return metrics
 
 
 class TrainValStage(Stage):
@@ -213,21 +255,21 @@ class TrainValStage(Stage):
         super().__init__()
         self.is_train = True
 
+    # -------------------------------------------------------- overridables
+
     def train_dataset(self):
-        train_ds = self.pipeline.datasets.get('train')
-        if train_ds is None:
-            raise ValueError(
-                'No "train" dataset found in pipeline. Use register_dataset("train", ...) to register a dataset.'
-            )
-        return train_ds
+        return self._lookup_dataset('train')
 
     def val_dataset(self):
-        val_ds = self.pipeline.datasets.get('val')
-        if val_ds is None:
+        return self._lookup_dataset('val')
+
+    def _lookup_dataset(self, key: str):
+        ds = self.pipeline.datasets.get(key)
+        if ds is None:
             raise ValueError(
-                'No "val" dataset found in pipeline. Use register_dataset("val", ...) to register a dataset.'
+                f'No "{key}" dataset found in pipeline. Use register_dataset("{key}", ...) to register a dataset.'
             )
-        return val_ds
+        return ds
 
     def optimizers(self):
         return self.pipeline.optimizers.values()
@@ -245,10 +287,6 @@ class TrainValStage(Stage):
         """Max gradient norm; 0 disables clipping."""
         return 0.0
 
-    def run_epoch(self):
-        self.train_epoch()
-        self.val_epoch()
-
     def step(self, batch) -> torch.Tensor:
         raise NotImplementedError()
 
@@ -258,6 +296,8 @@ class TrainValStage(Stage):
     def val_step(self, batch):
         return self.step(batch)
 
+    # ------------------------------------------------------------ hot path
+
     def zero_grad(self):
         for optimizer in self.optimizers():
             optimizer.zero_grad()
@@ -266,11 +306,19 @@ class TrainValStage(Stage):
         max_norm = self.gradient_clip()
         for optimizer in self.optimizers():
             if isinstance(optimizer, FlatOptimizer):
-                # fused device-side global-norm clip, no host sync
+                # fused device-side global-norm clip, no host sync;
+                # norm_scale inside accounts for the rank-summed grads
                 optimizer.clip_grad_norm_(max_norm)
             else:
                 for group in optimizer.param_groups:
                     torch.nn.utils.clip_grad_norm_(group['params'], max_norm)
+
+    def _grad_sync(self):
+        """Flat replicas need an explicit all-reduce (DDP syncs inside
+        backward)."""
+        for model in self.pipeline.models.values():
+            if hasattr(model, 'grad_sync'):
+                model.grad_sync()
 
     def optimize(self, loss):
         loss.backward()
@@ -280,46 +328,54 @@ class TrainValStage(Stage):
         for optimizer in self.optimizers():
             optimizer.step()
 
-    def _grad_sync(self):
-        """Flat replicas need an explicit all-reduce (DDP syncs inside
-        backward)."""
-        for model in self.pipeline.models.values():
-            if hasattr(model, 'grad_sync'):
-                model.grad_sync()
-
     def train_batch(self, batch):
         """One full training step: zero_grad -> forward -> backward ->
         (grad sync) -> optimizer, plus the standard step metrics. The
         benchmark driver calls this directly."""
-        step_start_time = time.perf_counter_ns()
+        t0 = time.perf_counter_ns()
         with roctx_range('train_batch'):
             self.zero_grad()
             loss = self.train_step(batch)
             self.optimize(loss)
-        step_end_time = time.perf_counter_ns()
+        elapsed_ms = (time.perf_counter_ns() - t0) / 1e6
 
         self.track_reduce(self.loss_metric_name(), loss)
-        self.track_reduce('misc/total_train_batches', torch.tensor(1), reduction=Reduction.SUM, prefixed=False)
+        self._count_batch('train')
+        self.track_reduce('misc/step_time_ms', torch.tensor(elapsed_ms), prefixed=False)
+
+    def _count_batch(self, split: str):
+        one = torch.tensor(1)
+        self.track_reduce(f'misc/total_{split}_batches', one, reduction=Reduction.SUM, prefixed=False)
         self.track_reduce(
-            'misc/worker_train_batches',
-            torch.tensor(1),
+            f'misc/worker_{split}_batches',
+            one,
             reduction=Reduction.SUM,
             reduce_globally=False,
             prefixed=False,
         )
-        self.track_reduce('misc/step_time_ms', torch.tensor(step_end_time - step_start_time) / 1e6, prefixed=False)
+
+    # --------------------------------------------------------- epoch loops
+
+    def run_epoch(self):
+        self.train_epoch()
+        self.val_epoch()
+
+    def _advance_sampler_epoch(self, loader):
+        """Tell a DataLoader's DistributedSampler (or an epoch-aware
+        dataset) which epoch this is, so shuffles differ per epoch."""
+        sampler = getattr(loader, 'sampler', None)
+        if sampler is not None and hasattr(sampler, 'set_epoch'):
+            sampler.set_epoch(self.current_epoch)
+        elif hasattr(loader, 'set_epoch'):
+            loader.set_epoch(self.current_epoch)
 
     def train_epoch(self):
         self.is_train = True
         self.metric_prefix = self.train_metric_prefix()
 
-        train_ds = self.train_dataset()
-        if hasattr(train_ds, 'sampler') and hasattr(train_ds.sampler, 'set_epoch'):
-            train_ds.sampler.set_epoch(self.current_epoch)
-        elif hasattr(train_ds, 'set_epoch'):
-            train_ds.set_epoch(self.current_epoch)
-
-        for batch in train_ds:
+        loader = self.train_dataset()
+        self._advance_sampler_epoch(loader)
+        for batch in loader:
             self.train_batch(batch)
 
         for name, scheduler in self.pipeline.schedulers.items():
@@ -334,14 +390,7 @@ class TrainValStage(Stage):
         for batch in self.val_dataset():
             loss = self.val_step(batch)
             self.track_reduce(self.loss_metric_name(), loss)
-            self.track_reduce('misc/total_val_batches', torch.tensor(1), reduction=Reduction.SUM, prefixed=False)
-            self.track_reduce(
-                'misc/worker_val_batches',
-                torch.tensor(1),
-                reduction=Reduction.SUM,
-                reduce_globally=False,
-                prefixed=False,
-            )
+            self._count_batch('val')
 
     def table_columns(self):
         columns = super().table_columns()

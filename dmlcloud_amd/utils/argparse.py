@@ -1,27 +1,50 @@
-"""Argparse helper for Enum-valued CLI flags.
+"""CLI helpers: Enum-valued argparse flags.
 
-Capability parity with reference dmlcloud/util/argparse.py:5-31.
+`parser.add_argument('--mode', type=Mode, action=EnumAction)` accepts the
+enum's *values* on the command line and stores the enum *member*
+(capability parity with reference dmlcloud/util/argparse.py:5-31).
+Supports repeated flags via nargs (stores a list of members) and derives
+the choices list from the enum automatically.
 """
 
 import argparse
-import enum
+from enum import Enum
+from typing import Optional, Sequence, Union
+
+__all__ = ['EnumAction']
 
 
 class EnumAction(argparse.Action):
-    """Argparse action that parses string choices into an Enum."""
+    """Store an Enum member parsed from its string value.
 
-    def __init__(self, **kwargs):
-        enum_type = kwargs.pop('type', None)
+    The ``type`` keyword carries the Enum class (argparse never calls it
+    as a converter — this action consumes it instead). ``choices``
+    defaults to every member's value.
+    """
 
-        if enum_type is None:
-            raise ValueError('type must be assigned an Enum when using EnumAction')
-        if not issubclass(enum_type, enum.Enum):
-            raise TypeError('type must be an Enum when using EnumAction')
+    def __init__(self, option_strings, dest, type=None, choices=None, **kwargs):
+        if type is None:
+            raise ValueError('EnumAction needs type=<Enum subclass>')
+        try:
+            is_enum = issubclass(type, Enum)
+        except TypeError:
+            is_enum = False
+        if not is_enum:
+            raise TypeError(f'EnumAction type must be an Enum subclass, got {type!r}')
+        self._enum_cls = type
+        if choices is None:
+            choices = tuple(member.value for member in self._enum_cls)
+        super().__init__(option_strings, dest, choices=choices, **kwargs)
 
-        kwargs.setdefault('choices', tuple(e.value for e in enum_type))
-
-        super().__init__(**kwargs)
-        self._enum = enum_type
-
-    def __call__(self, parser, namespace, values, option_string=None):
-        setattr(namespace, self.dest, self._enum(values))
+    def __call__(
+        self,
+        parser: argparse.ArgumentParser,
+        namespace: argparse.Namespace,
+        values: Union[str, Sequence],
+        option_string: Optional[str] = None,
+    ):
+        if isinstance(values, (list, tuple)):
+            parsed = [self._enum_cls(v) for v in values]
+        else:
+            parsed = self._enum_cls(values)
+        setattr(namespace, self.dest, parsed)

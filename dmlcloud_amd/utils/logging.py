@@ -1,9 +1,19 @@
-"""Logging, IO redirection and startup diagnostics.
+"""Logging, stdio capture, and the startup diagnostics report.
 
-Capability parity with reference dmlcloud/util/logging.py:18-173, with the
-GPU probes swapped for the ROCm stack: `rocm-smi`/`amd-smi` instead of
-`nvidia-smi -L` (reference logging.py:146), `torch.version.hip` instead of
-`torch.version.cuda`, and the ROCm version read from /opt/rocm.
+Three jobs (capability parity with reference dmlcloud/util/logging.py:
+18-173, reshaped for the ROCm stack):
+
+1. ``IORedirector`` — capture everything written to stdout/stderr into
+   the run's ``log.txt`` while still reaching the terminal. Implemented
+   as a single tee proxy class used for both streams.
+2. Rank-aware handler setup for the ``'dmlcloud'`` logger: rank 0 is
+   chatty (INFO), other ranks only surface problems (WARNING); records
+   below WARNING go to stdout, the rest to stderr.
+3. ``general_diagnostics`` — a provenance block logged at startup:
+   process identity, git state, distributed backend, the ROCm/HIP
+   stack (``amd-smi`` / ``rocm-smi`` / ``torch.version.hip`` — the
+   CUDA probes of the reference have no ROCm analog and are replaced,
+   reference logging.py:146,155), package versions, and SLURM context.
 """
 
 import io
@@ -13,99 +23,128 @@ import subprocess
 import sys
 from datetime import datetime
 from pathlib import Path
+from typing import List, Optional, TextIO
 
 import torch
 import torch.distributed as dist
 
-from . import slurm
 from .git import git_hash
-from .thirdparty import ML_MODULES, is_imported, try_get_version
+from .slurm import slurm_job_id, slurm_step_id
+from .thirdparty import installed_versions
+
+__all__ = [
+    'IORedirector',
+    'DevNullIO',
+    'add_log_handlers',
+    'flush_log_handlers',
+    'experiment_header',
+    'general_diagnostics',
+]
+
+
+class _Tee:
+    """File-like proxy that writes to a live terminal stream and a file."""
+
+    def __init__(self, terminal: TextIO, sink: TextIO):
+        self._terminal = terminal
+        self._sink = sink
+
+    def write(self, text):
+        self._sink.write(text)
+        return self._terminal.write(text)
+
+    def flush(self):
+        self._sink.flush()
+        self._terminal.flush()
+
+    def __getattr__(self, attr):
+        # isatty, encoding, fileno, ... — delegate to the real terminal
+        return getattr(self._terminal, attr)
 
 
 class IORedirector:
-    """Tees stdout/stderr into a log file while preserving the originals."""
+    """Mirror stdout and stderr into ``log_file`` (append mode).
 
-    class Stdout:
-        def __init__(self, parent):
-            self.parent = parent
-
-        def write(self, data):
-            self.parent.file.write(data)
-            self.parent.stdout.write(data)
-
-        def flush(self):
-            self.parent.file.flush()
-            self.parent.stdout.flush()
-
-    class Stderr:
-        def __init__(self, parent):
-            self.parent = parent
-
-        def write(self, data):
-            self.parent.file.write(data)
-            self.parent.stderr.write(data)
-
-        def flush(self):
-            self.parent.file.flush()
-            self.parent.stderr.flush()
+    ``install()`` swaps ``sys.stdout``/``sys.stderr`` for tee proxies;
+    ``uninstall()`` restores the originals and closes the file. Also
+    usable as a context manager. Installing twice is a no-op.
+    """
 
     def __init__(self, log_file: Path):
-        self.path = log_file
-        self.file = None
-        self.stdout = None
-        self.stderr = None
+        self.path = Path(log_file)
+        self._log: Optional[TextIO] = None
+        self._saved = None
+
+    @property
+    def active(self) -> bool:
+        return self._log is not None
 
     def install(self):
-        if self.file is not None:
+        if self.active:
             return
-        self.file = open(self.path, 'a')
-        self.stdout = sys.stdout
-        self.stderr = sys.stderr
-        self.stdout.flush()
-        self.stderr.flush()
-        sys.stdout = self.Stdout(self)
-        sys.stderr = self.Stderr(self)
+        self._log = open(self.path, 'a', buffering=1)
+        self._saved = (sys.stdout, sys.stderr)
+        for stream in self._saved:
+            stream.flush()
+        sys.stdout = _Tee(self._saved[0], self._log)
+        sys.stderr = _Tee(self._saved[1], self._log)
 
     def uninstall(self):
-        self.stdout.flush()
-        self.stderr.flush()
-        sys.stdout = self.stdout
-        sys.stderr = self.stderr
-        self.file.close()
+        if not self.active:
+            return
+        sys.stdout.flush()
+        sys.stderr.flush()
+        sys.stdout, sys.stderr = self._saved
+        self._log.close()
+        self._log = None
+        self._saved = None
 
     def __enter__(self):
         self.install()
         return self
 
-    def __exit__(self, exc_type, exc_value, traceback):
+    def __exit__(self, *exc_info):
         self.uninstall()
+
+    # kept for callers that used the attribute names of the old API
+    @property
+    def file(self):
+        return self._log
 
 
 class DevNullIO(io.TextIOBase):
-    """Sink that ignores all writes (non-root progress tables)."""
+    """Write sink that discards everything (non-root progress tables)."""
 
-    def write(self, msg):
-        pass
+    def write(self, text):
+        return len(text) if text else 0
+
+    def writable(self):
+        return True
+
+
+def _is_root_rank() -> bool:
+    return not dist.is_initialized() or dist.get_rank() == 0
 
 
 def add_log_handlers(logger: logging.Logger):
-    """Root rank logs at INFO, others at WARNING; <WARNING goes to stdout,
-    >=WARNING to stderr."""
+    """Attach the rank-aware stdout/stderr handler pair (idempotent).
+
+    Rank 0: INFO and up. Other ranks: WARNING and up. Severity below
+    WARNING renders to stdout, WARNING+ to stderr — so redirected logs
+    interleave correctly with print() output.
+    """
     if logger.hasHandlers():
         return
+    logger.setLevel(logging.INFO if _is_root_rank() else logging.WARNING)
 
-    logger.setLevel(logging.INFO if dist.get_rank() == 0 else logging.WARNING)
+    info_handler = logging.StreamHandler(sys.stdout)
+    info_handler.setLevel(logging.DEBUG)
+    info_handler.addFilter(lambda rec: rec.levelno < logging.WARNING)
+    logger.addHandler(info_handler)
 
-    stdout_handler = logging.StreamHandler(sys.stdout)
-    stdout_handler.setLevel(logging.DEBUG)
-    stdout_handler.addFilter(lambda record: record.levelno < logging.WARNING)
-    stdout_handler.setFormatter(logging.Formatter())
-    logger.addHandler(stdout_handler)
-
-    stderr_handler = logging.StreamHandler()
-    stderr_handler.setLevel(logging.WARNING)
-    stderr_handler.setFormatter(logging.Formatter())
-    logger.addHandler(stderr_handler)
+    problem_handler = logging.StreamHandler(sys.stderr)
+    problem_handler.setLevel(logging.WARNING)
+    logger.addHandler(problem_handler)
 
 
 def flush_log_handlers(logger: logging.Logger):
@@ -114,70 +153,93 @@ def flush_log_handlers(logger: logging.Logger):
 
 
 def experiment_header(name, checkpoint_dir, date: datetime) -> str:
-    msg = f'...............  Experiment: {name if name else "N/A"}  ...............\n'
-    msg += f'- Date: {date}\n'
-    msg += f'- Checkpoint Dir: {checkpoint_dir if checkpoint_dir else "N/A"}\n'
-    msg += f'- Training on {dist.get_world_size()} GPUs\n'
-    return msg
+    """Banner logged when a pipeline starts."""
+    world = dist.get_world_size() if dist.is_initialized() else 1
+    lines = [
+        f'...............  Experiment: {name or "N/A"}  ...............',
+        f'- Date: {date}',
+        f'- Checkpoint Dir: {checkpoint_dir or "N/A"}',
+        f'- Training on {world} GPUs',
+    ]
+    return '\n'.join(lines) + '\n'
 
 
-def _rocm_version() -> str:
-    try:
-        return Path('/opt/rocm/.info/version').read_text().strip()
-    except (FileNotFoundError, OSError):
-        return 'N/A'
+# ------------------------------------------------------------- diagnostics
 
 
-def _gpu_listing() -> list:
-    """Enumerate GPUs via amd-smi/rocm-smi, falling back to torch."""
-    for cmd in (['amd-smi', 'list', '--csv'], ['rocm-smi', '--showproductname']):
+def _read_rocm_release() -> Optional[str]:
+    for candidate in ('/opt/rocm/.info/version', '/opt/rocm/.info/version-dev'):
         try:
-            proc = subprocess.run(cmd, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, timeout=30)
-            if proc.returncode == 0:
-                return proc.stdout.decode().splitlines()
+            return Path(candidate).read_text().strip()
+        except OSError:
+            continue
+    return None
+
+
+def _probe_gpus() -> List[str]:
+    """GPU inventory via the AMD SMI tools, torch as a fallback."""
+    smi_commands = (
+        ['amd-smi', 'list', '--csv'],
+        ['rocm-smi', '--showproductname'],
+    )
+    for command in smi_commands:
+        try:
+            result = subprocess.run(
+                command, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, timeout=30
+            )
         except (FileNotFoundError, subprocess.TimeoutExpired):
             continue
-    return [f'{torch.cuda.get_device_name(i)}' for i in range(torch.cuda.device_count())]
+        if result.returncode == 0:
+            return [ln for ln in result.stdout.decode().splitlines() if ln.strip()]
+    return [torch.cuda.get_device_name(i) for i in range(torch.cuda.device_count())]
+
+
+def _bullet_block(title: str, entries) -> str:
+    body = ''.join(f'    - {key}: {val}\n' for key, val in entries)
+    return f'* {title}:\n{body}'
 
 
 def general_diagnostics() -> str:
+    """Multi-line provenance report for the start-of-run log."""
     import dmlcloud_amd
 
-    msg = '* GENERAL:\n'
-    msg += f'    - argv: {sys.argv}\n'
-    msg += f'    - cwd: {Path.cwd()}\n'
-    msg += f'    - host (root): {os.environ.get("HOSTNAME")}\n'
-    msg += f'    - user: {os.environ.get("USER")}\n'
-    msg += f'    - git-hash: {git_hash()}\n'
-    msg += f'    - conda-env: {os.environ.get("CONDA_DEFAULT_ENV", "N/A")}\n'
-    msg += f'    - sys-prefix: {sys.prefix}\n'
-    msg += f'    - backend: {dist.get_backend()}\n'
-    msg += f'    - gpu available: {torch.cuda.is_available()}\n'
+    general = [
+        ('argv', sys.argv),
+        ('cwd', Path.cwd()),
+        ('host (root)', os.environ.get('HOSTNAME')),
+        ('user', os.environ.get('USER')),
+        ('git-hash', git_hash()),
+        ('conda-env', os.environ.get('CONDA_DEFAULT_ENV', 'N/A')),
+        ('sys-prefix', sys.prefix),
+        ('backend', dist.get_backend() if dist.is_initialized() else 'N/A'),
+        ('gpu available', torch.cuda.is_available()),
+    ]
+    report = _bullet_block('GENERAL', general)
 
     if torch.cuda.is_available():
-        msg += '* GPUs (root):\n'
-        for line in _gpu_listing():
-            if line.strip():
-                msg += f'    - {line}\n'
+        report += _bullet_block('GPUs (root)', [('gpu', g) for g in _probe_gpus()]).replace(
+            '- gpu: ', '- '
+        )
 
-    msg += '* VERSIONS:\n'
-    msg += f'    - python: {sys.version}\n'
-    msg += f'    - dmlcloud_amd: {dmlcloud_amd.__version__}\n'
-    msg += f'    - hip: {torch.version.hip}\n'
-    msg += f'    - rocm: {_rocm_version()}\n'
+    versions = [
+        ('python', sys.version),
+        ('dmlcloud_amd', dmlcloud_amd.__version__),
+        ('hip', torch.version.hip),
+        ('rocm', _read_rocm_release() or 'N/A'),
+    ]
+    versions += sorted(installed_versions().items())
+    report += _bullet_block('VERSIONS', versions)
 
-    for module_name in ML_MODULES:
-        if is_imported(module_name):
-            msg += f'    - {module_name}: {try_get_version(module_name)}\n'
+    if slurm_job_id() is not None:
+        slurm_vars = [('SLURM_JOB_ID', slurm_job_id()), ('SLURM_STEP_ID', slurm_step_id())]
+        for env_name in (
+            'SLURM_STEP_NODELIST',
+            'SLURM_TASKS_PER_NODE',
+            'SLURM_STEP_GPUS',
+            'SLURM_GPUS_ON_NODE',
+            'SLURM_CPUS_PER_TASK',
+        ):
+            slurm_vars.append((env_name, os.environ.get(env_name)))
+        report += _bullet_block('SLURM', slurm_vars)
 
-    if 'SLURM_JOB_ID' in os.environ:
-        msg += '* SLURM:\n'
-        msg += f'    - SLURM_JOB_ID = {slurm.slurm_job_id()}\n'
-        msg += f'    - SLURM_STEP_ID = {slurm.slurm_step_id()}\n'
-        msg += f'    - SLURM_STEP_NODELIST = {os.environ.get("SLURM_STEP_NODELIST")}\n'
-        msg += f'    - SLURM_TASKS_PER_NODE = {os.environ.get("SLURM_TASKS_PER_NODE")}\n'
-        msg += f'    - SLURM_STEP_GPUS = {os.environ.get("SLURM_STEP_GPUS")}\n'
-        msg += f'    - SLURM_GPUS_ON_NODE = {os.environ.get("SLURM_GPUS_ON_NODE")}\n'
-        msg += f'    - SLURM_CPUS_PER_TASK = {os.environ.get("SLURM_CPUS_PER_TASK")}'
-
-    return msg
+    return report.rstrip('\n')
