@@ -59,6 +59,14 @@ def parse_args():
     p.add_argument('--tunableop', action='store_true', help='enable PyTorch TunableOp GEMM tuning (hipBLASLt)')
     p.add_argument('--no-channels-last', action='store_true', help='disable NHWC memory format (resnet50)')
     p.add_argument('--seq-len', type=int, default=1024, help='gpt2 sequence length')
+    p.add_argument(
+        '--metric-stress',
+        type=int,
+        default=0,
+        metavar='N',
+        help='track N distributed reduced metrics per step (BASELINE config #5: '
+        'ResNet-50 bs=8192 reducer stress — combine with --model resnet50 --batch-size 8192)',
+    )
     return p.parse_args()
 
 
@@ -181,6 +189,13 @@ class BenchStage(TrainValStage):
             loss = self.train_step((self.static_batch, self.static_labels))
         self.optimize(loss)
         self.track_reduce(self.loss_metric_name(), loss)
+        if self.args.metric_stress:
+            # BASELINE config #5: N device-resident reducer accumulations
+            # per step (one metric_reduce_into kernel chain each; the
+            # epoch-end RCCL collectives fuse into one per reduction op)
+            rotations = (Reduction.MEAN, Reduction.SUM, Reduction.MIN, Reduction.MAX)
+            for k in range(self.args.metric_stress):
+                self.track_reduce(f'stress_{k}', loss, reduction=rotations[k % 4], prefixed=False)
 
     def load_batch(self, i):
         """Stage pool batch i into the static buffers (device-side copy)."""
@@ -213,6 +228,11 @@ def main():
     else:
         device = torch.device('cpu')
 
+    if world > 1 and device.type == 'cuda':
+        # establish the RCCL communicator before any timed/captured work
+        dist.all_reduce(torch.zeros(1, device=device))
+        torch.cuda.synchronize()
+
     if args.batch_size is None:
         args.batch_size = {'mnist': 16384, 'resnet50': 512, 'gpt2': 64}[args.model]
 
@@ -227,7 +247,10 @@ def main():
 
     if use_graph:
         stage.load_batch(0)
-        graphed = GraphedStep(stage.core_step, warmup=3)
+        # validate=True: a capture that replays incorrectly (e.g. an RCCL
+        # collective that cannot replay at this world size) falls back to
+        # eager — measured, not assumed (the JSON reports hipgraph: false)
+        graphed = GraphedStep(stage.core_step, warmup=3, validate=True)
         graphed.initialize()
 
         def run_step(i):
@@ -299,6 +322,7 @@ def main():
                 'impl': args.impl,
                 'hipgraph': bool(use_graph and getattr(graphed, 'captured', False)) if use_graph else False,
                 **({'seq_len': args.seq_len} if args.model == 'gpt2' else {}),
+                **({'metric_stress': args.metric_stress} if args.metric_stress else {}),
             },
         }
         print(json.dumps(result))

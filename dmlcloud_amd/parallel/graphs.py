@@ -31,10 +31,19 @@ class GraphedStep:
         gs()                     # replay (or eager fallback)
     """
 
-    def __init__(self, step_fn: Callable[[], None], warmup: int = 3, enabled: bool = True):
+    def __init__(
+        self, step_fn: Callable[[], None], warmup: int = 3, enabled: bool = True, validate: bool = False
+    ):
+        """validate=True replays the captured graph once (and synchronizes)
+        inside initialize() so replay-time failures — e.g. an RCCL
+        collective that captured but cannot replay at this world size —
+        also trigger the eager fallback. The validation replay advances
+        model/optimizer state by one step, so enable it only when
+        initialize() runs inside a warmup phase (the benchmark does)."""
         self.step_fn = step_fn
         self.warmup = warmup
         self.enabled = enabled and torch.cuda.is_available()
+        self.validate = validate
         self.graph: Optional[torch.cuda.CUDAGraph] = None
         self._initialized = False
 
@@ -56,6 +65,9 @@ class GraphedStep:
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 self.step_fn()
+            if self.validate:
+                graph.replay()
+                torch.cuda.synchronize()
             self.graph = graph
         except Exception as e:  # pragma: no cover - depends on runtime support
             logger.warning(f'hipGraph capture failed ({e!r}); falling back to eager stepping')

@@ -343,6 +343,14 @@ class TrainingPipeline:
             warnings.warn('No GPU available. Running on CPU.')
             self.device = torch.device('cpu')
 
+        if self.device.type == 'cuda' and dist.get_world_size() > 1:
+            # First collective initializes the RCCL communicator lazily;
+            # do it NOW so communicator setup (xGMI ring/tree discovery)
+            # never lands inside the timed training path or a later
+            # hipGraph capture region.
+            dist.all_reduce(torch.zeros(1, device=self.device))
+            torch.cuda.synchronize()
+
         # prevent checkpoint dir creation before all ranks searched for it
         self.barrier(timeout=10 * 60)
         if self.checkpointing_enabled:
