@@ -101,7 +101,19 @@ class _FakeXr:
         return self
 
 
+def _rank_chunks(n, chunk_size, world_size, **kwargs):
+    """[rank][chunk] element lists for every rank of a _FakeXr(n)."""
+    ds = _FakeXr(n)
+    return [
+        [c.data for c in sharded_xr_dataset(ds, 'time', chunk_size, rank=r, world_size=world_size, **kwargs)]
+        for r in range(world_size)
+    ]
+
+
 class TestShardedXr:
+    """Chunked sharding case matrix (reference test/test_data.py:57-170,
+    365-441 — same cases, duck-typed stand-in instead of xarray)."""
+
     def test_basic(self):
         ds = _FakeXr(20)
         chunks = list(sharded_xr_dataset(ds, 'time', 5, rank=0, world_size=2))
@@ -119,6 +131,114 @@ class TestShardedXr:
             for c in sharded_xr_dataset(ds, 'time', 5, rank=r, world_size=3):
                 seen.extend(c.data)
         assert sorted(seen) == list(range(30))
+
+    def test_w3_exact(self):
+        """100 elements / chunk 15 / 3 ranks: 6 full chunks round-robin."""
+        per_rank = _rank_chunks(100, 15, 3)
+        assert [len(cs) for cs in per_rank] == [2, 2, 2]
+        assert per_rank[0][0] == list(range(0, 15))
+        assert per_rank[1][0] == list(range(15, 30))
+        assert per_rank[2][0] == list(range(30, 45))
+        assert per_rank[0][1] == list(range(45, 60))
+        assert per_rank[1][1] == list(range(60, 75))
+        assert per_rank[2][1] == list(range(75, 90))
+
+    def test_uneven_shards(self):
+        """even_shards=False keeps the 5th chunk: rank 2 gets one chunk."""
+        per_rank = _rank_chunks(100, 20, 3, even_shards=False)
+        assert [len(cs) for cs in per_rank] == [2, 2, 1]
+        assert per_rank[0] == [list(range(0, 20)), list(range(60, 80))]
+        assert per_rank[1] == [list(range(20, 40)), list(range(80, 100))]
+        assert per_rank[2] == [list(range(40, 60))]
+
+    def test_unequal_chunks(self):
+        """equal_chunks=False emits a short trailing chunk."""
+        per_rank = _rank_chunks(110, 20, 3, equal_chunks=False)
+        assert [len(cs) for cs in per_rank] == [2, 2, 2]
+        assert per_rank[2][1] == list(range(100, 110))  # size 10
+        assert per_rank[0][1] == list(range(60, 80))
+
+    def test_shuffled(self):
+        """Shuffled chunks: every chunk stays contiguous, the union covers
+        the 6 full chunks, the order differs from unshuffled."""
+        per_rank = _rank_chunks(100, 15, 3, shuffle=True, seed=0)
+        flattened = [x for cs in per_rank for c in cs for x in c]
+        assert sorted(flattened) == list(range(90))
+        assert flattened != list(range(90))
+        for cs in per_rank:
+            for chunk in cs:
+                assert chunk == list(range(chunk[0], chunk[-1] + 1))
+
+    def test_overlap_w3_exact(self):
+        """overlap=5 extends every window right by 5 elements."""
+        per_rank = _rank_chunks(100, 15, 3, chunk_overlap=5)
+        assert [len(cs) for cs in per_rank] == [2, 2, 2]
+        assert per_rank[0][0] == list(range(0, 20))
+        assert per_rank[1][0] == list(range(15, 35))
+        assert per_rank[2][0] == list(range(30, 50))
+        assert per_rank[0][1] == list(range(45, 65))
+        assert per_rank[1][1] == list(range(60, 80))
+        assert per_rank[2][1] == list(range(75, 95))
+
+    def test_overlap_unequal_uneven(self):
+        """All three flags at once: 7 chunks, rank 0 also gets the clipped
+        tail window (90, 110) -> [90:100]."""
+        per_rank = _rank_chunks(100, 15, 3, chunk_overlap=5, even_shards=False, equal_chunks=False)
+        assert [len(cs) for cs in per_rank] == [3, 2, 2]
+        assert per_rank[0][0] == list(range(0, 20))
+        assert per_rank[1][0] == list(range(15, 35))
+        assert per_rank[2][0] == list(range(30, 50))
+        assert per_rank[0][1] == list(range(45, 65))
+        assert per_rank[1][1] == list(range(60, 80))
+        assert per_rank[2][1] == list(range(75, 95))
+        assert per_rank[0][2] == list(range(90, 100))
+
+
+class _FlattenChunks(torch.utils.data.IterableDataset):
+    """Yield the individual elements of each chunk a wrapped chunked
+    dataset produces (the reference's _Unzip, for exact-order asserts)."""
+
+    def __init__(self, chunked):
+        self.chunked = chunked
+
+    def __iter__(self):
+        for chunk in self.chunked:
+            yield from chunk.data
+
+
+def _loader_order(n, chunk_size, rank, world_size, num_workers=2):
+    from dmlcloud_amd.data import ShardedXrDataset
+
+    chunked = ShardedXrDataset(_FakeXr(n), 'time', chunk_size, rank=rank, world_size=world_size)
+    loader = DataLoader(_FlattenChunks(chunked), batch_size=1, num_workers=num_workers, prefetch_factor=1)
+    return [int(b) for b in loader]
+
+
+class TestShardedXrWorkerInterleaving:
+    """Exact element order through DataLoader workers — the behavioral
+    contract of worker-id folding (reference test/test_data.py:171-363)."""
+
+    def test_two_workers_world1(self):
+        # effective world 2: worker0 owns chunks 0,2,4; worker1 owns 1,3,5.
+        # batch_size=1 round-robins the workers element by element.
+        out = _loader_order(100, 15, rank=0, world_size=1)
+        expected = []
+        for lo_a, lo_b in ((0, 15), (30, 45), (60, 75)):
+            for i in range(15):
+                expected += [lo_a + i, lo_b + i]
+        assert out == expected
+
+    def test_two_workers_world2_rank0(self):
+        # effective world 4, 6 chunks, even_shards drops 2: rank0 gets
+        # chunks 0 (worker0) and 1 (worker1), interleaved per element
+        out = _loader_order(100, 15, rank=0, world_size=2)
+        expected = [x for i in range(15) for x in (i, 15 + i)]
+        assert out == expected
+
+    def test_two_workers_world2_rank1(self):
+        out = _loader_order(100, 15, rank=1, world_size=2)
+        expected = [x for i in range(15) for x in (30 + i, 45 + i)]
+        assert out == expected
 
 
 class TestShardedSequenceDataset:
