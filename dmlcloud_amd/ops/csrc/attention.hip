@@ -31,13 +31,21 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 constexpr int kAttnD = 64; // head dim (v1: fixed)
 constexpr int kQT = 16; // query rows per wave
-constexpr int kKT = 32; // keys per kv tile
-constexpr int kPStride = 40; // P_lds row stride in bf16 (16B-aligned rows)
+constexpr int kKT = 32; // keys per kv tile (backward kernels)
+constexpr int kPStride = 40; // P_lds row stride in bf16 (16B-aligned rows, backward)
 constexpr int kWavesPerBlock = 4;
 // K/V LDS row stride in bf16: multiple of 8 (16B-aligned vector rows) with
 // (stride/2 dwords, 64 banks) gcd = 4 so 16 simultaneous row reads at one
 // column offset span 16 distinct banks (rows at stride 88 bf16 = 44 dwords)
 constexpr int kKVStride = 88;
+// Forward v7 uses 64-key tiles: keys/softmax-round doubles, so the
+// per-round fixed costs (max/sum cross-lane reductions, running-stat
+// updates, O-accumulator rescale, P LDS round-trip waits, the block-wide
+// staging barrier) halve per key while the MFMA count per key stays
+// identical. 16 rows at stride 72 bf16 = 36 dwords hit 16 distinct banks
+// (gcd(36,64)=4).
+constexpr int kKTF = 64; // fwd: keys per kv tile
+constexpr int kPStrideF = 72; // fwd: P_lds row stride in bf16
 
 // Tiled 32x64 LDS image index for ds_read_b64_tr_b16 consumption
 // (tools/tr_probe.hip): elem[(col/16*2 + row/4%2)*256 + row/8*64 +
@@ -71,24 +79,27 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
     float* __restrict__ lse, int BH, int H, int N, float scale, bool causal, Strides sq,
     Strides sk, Strides sv) {
-  __shared__ __hip_bfloat16 p_lds_all[kWavesPerBlock][kQT][kPStride];
-  __shared__ __hip_bfloat16 k_lds[2][kKT][kKVStride];
-  // V lives in a TILED image read by ds_read_b64_tr_b16 (empirically
+  __shared__ __hip_bfloat16 p_lds_all[kWavesPerBlock][kQT][kPStrideF];
+  __shared__ __hip_bfloat16 k_lds[2][kKTF][kKVStride];
+  // V lives in TILED images read by ds_read_b64_tr_b16 (empirically
   // probed semantics, tools/tr_probe.hip: 16 contiguous per-lane 8-byte
-  // addresses cover a 64-element region transposed as 4x16):
-  //   elem[(db*2+half)*256 + g*64 + jj*16 + c] = V[8g+4half+jj][16db+c]
-  __shared__ __hip_bfloat16 v_tr[2][2 * kKT * kAttnD / 2]; // [2][2048]
+  // addresses cover a 64-element region transposed as 4x16). A 64-key
+  // tile is TWO consecutive 32x64 images (key chunk c at offset c*2048):
+  //   elem[c*2048 + (db*2+half)*256 + g*64 + jj*16 + cc] = V[32c+8g+4half+jj][16db+cc]
+  __shared__ __hip_bfloat16 v_tr[2][2 * kKTF * kAttnD / 2]; // [2][4096]
 
   const int lane = threadIdx.x & (kWave - 1);
   const int wave = threadIdx.x / kWave;
   const int row16 = lane & 15; // q index inside the tile (and key row for K frags)
   const int grp = lane >> 4; // 16-lane group 0..3
-  __hip_bfloat16(*p_lds)[kPStride] = p_lds_all[wave];
+  __hip_bfloat16(*p_lds)[kPStrideF] = p_lds_all[wave];
 
-  // staging coords: thread t loads row t/8, bf16x8 chunk t%8 of a 32x64 tile
+  // staging coords: 64x64 tile = 512 bf16x8 pieces, 2 per thread:
+  // (st_row, st_col) and (st_row + 32, st_col)
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 8;
-  // tiled V image offset for this thread's (st_row, st_col) bf16x8 piece
+  // tiled V image offset for the (st_row, st_col) piece; the second
+  // piece lands at the same offset in the second 32x64 image (+2048)
   const int st_vt = ((st_col >> 4) * 2 + ((st_row >> 2) & 1)) * 256 + (st_row >> 3) * 64 +
                     (st_row & 3) * 16 + (st_col & 15);
 
@@ -133,45 +144,56 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
     // loads stay in flight through the current tile's compute; their
     // ds_write targets the other buffer just before the single barrier.
     *(bf16x8*)(&k_lds[0][st_row][st_col]) = *(const bf16x8*)(kp + (int64_t)st_row * sk.r + st_col);
+    *(bf16x8*)(&k_lds[0][st_row + 32][st_col]) =
+        *(const bf16x8*)(kp + (int64_t)(st_row + 32) * sk.r + st_col);
     *(bf16x8*)(&v_tr[0][st_vt]) = *(const bf16x8*)(vp + (int64_t)st_row * sv.r + st_col);
+    *(bf16x8*)(&v_tr[0][2048 + st_vt]) =
+        *(const bf16x8*)(vp + (int64_t)(st_row + 32) * sv.r + st_col);
     __syncthreads();
 
-    const int ntiles = (kv_end_block + kKT - 1) / kKT;
+    const int ntiles = (kv_end_block + kKTF - 1) / kKTF;
     for (int jt = 0; jt < ntiles; ++jt) {
-      const int j0 = jt * kKT;
+      const int j0 = jt * kKTF;
       const int buf = jt & 1;
-      bf16x8 knext, vnext;
+      bf16x8 knext0, knext1, vnext0, vnext1;
       const bool has_next = jt + 1 < ntiles;
       if (has_next) {
-        knext = *(const bf16x8*)(kp + (int64_t)(j0 + kKT + st_row) * sk.r + st_col);
-        vnext = *(const bf16x8*)(vp + (int64_t)(j0 + kKT + st_row) * sv.r + st_col);
+        knext0 = *(const bf16x8*)(kp + (int64_t)(j0 + kKTF + st_row) * sk.r + st_col);
+        knext1 = *(const bf16x8*)(kp + (int64_t)(j0 + kKTF + 32 + st_row) * sk.r + st_col);
+        vnext0 = *(const bf16x8*)(vp + (int64_t)(j0 + kKTF + st_row) * sv.r + st_col);
+        vnext1 = *(const bf16x8*)(vp + (int64_t)(j0 + kKTF + 32 + st_row) * sv.r + st_col);
       }
 
       if (valid && j0 < my_kv_end) {
-        // shared V fragments for both q sub-tiles: two hardware transpose
-        // reads per fragment instead of 8 scalar reads + packing
+        // shared V fragments for both q sub-tiles: [key chunk][D chunk],
+        // two hardware transpose reads per fragment
         typedef short short4v __attribute__((ext_vector_type(4)));
-        bf16x8 vf[4];
+        bf16x8 vf[2][4];
 #pragma unroll
-        for (int db = 0; db < 4; ++db) {
-          const short4v t0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (__attribute__((address_space(3))) short4v*)&v_tr[buf][(db * 2 + 0) * 256 +
-                                                                     lane * 4]);
-          const short4v t1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (__attribute__((address_space(3))) short4v*)&v_tr[buf][(db * 2 + 1) * 256 +
-                                                                     lane * 4]);
-          __builtin_memcpy(&vf[db], &t0, 8);
-          __builtin_memcpy((char*)&vf[db] + 8, &t1, 8);
+        for (int c = 0; c < 2; ++c) {
+#pragma unroll
+          for (int db = 0; db < 4; ++db) {
+            const short4v t0 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                (__attribute__((address_space(3))) short4v*)&v_tr[buf][c * 2048 +
+                                                                       (db * 2 + 0) * 256 +
+                                                                       lane * 4]);
+            const short4v t1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+                (__attribute__((address_space(3))) short4v*)&v_tr[buf][c * 2048 +
+                                                                       (db * 2 + 1) * 256 +
+                                                                       lane * 4]);
+            __builtin_memcpy(&vf[c][db], &t0, 8);
+            __builtin_memcpy((char*)&vf[c][db] + 8, &t1, 8);
+          }
         }
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
           const int qi0 = i0 + 16 * sub;
           if (causal && j0 >= qi0 + kQT) continue; // tile fully past diagonal
 
-          // ---- S^T = K Q^T for two 16-key halves ----
-          float sv[8]; // h*4 + r: key = j0 + 16h + 4grp + r, query = row16
+          // ---- S^T = K Q^T for four 16-key quarters ----
+          float sv[16]; // h*4 + r: key = j0 + 16h + 4grp + r, query = row16
 #pragma unroll
-          for (int h = 0; h < 2; ++h) {
+          for (int h = 0; h < 4; ++h) {
             f32x4 acc = {0, 0, 0, 0};
 #pragma unroll
             for (int c = 0; c < 2; ++c) {
@@ -193,7 +215,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
           // ---- online softmax (per q = row16; reduce across grp groups) ----
           float mt = sv[0];
 #pragma unroll
-          for (int x = 1; x < 8; ++x) mt = fmaxf(mt, sv[x]);
+          for (int x = 1; x < 16; ++x) mt = fmaxf(mt, sv[x]);
           mt = fmaxf(mt, __shfl_xor(mt, 16, kWave));
           mt = fmaxf(mt, __shfl_xor(mt, 32, kWave));
           const float m_new = fmaxf(m_run[sub], mt);
@@ -201,7 +223,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
 
           float ps = 0.0f;
 #pragma unroll
-          for (int x = 0; x < 8; ++x) {
+          for (int x = 0; x < 16; ++x) {
             sv[x] = __builtin_amdgcn_exp2f(sv[x] - m_new); // raw v_exp_f32 rate
             ps += sv[x];
           }
@@ -212,7 +234,7 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
 
           // ---- P^T -> per-wave LDS slice (4 keys per 8-byte write) ----
 #pragma unroll
-          for (int h = 0; h < 2; ++h) {
+          for (int h = 0; h < 4; ++h) {
             bf16x4 pw;
 #pragma unroll
             for (int r = 0; r < 4; ++r) pw[r] = (__bf16)sv[h * 4 + r];
@@ -224,19 +246,24 @@ __global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
           for (int r = 0; r < 4; ++r) a_o[r] = __shfl(alpha, 4 * grp + r, kWave);
 
           // ---- PV: A = P[q][key] from LDS, B = shared V fragments ----
-          const bf16x8 pf = *(const bf16x8*)(&p_lds[row16][8 * grp]);
+          const bf16x8 pf0 = *(const bf16x8*)(&p_lds[row16][8 * grp]);
+          const bf16x8 pf1 = *(const bf16x8*)(&p_lds[row16][32 + 8 * grp]);
 #pragma unroll
           for (int db = 0; db < 4; ++db) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) o_acc[sub][db][r] *= a_o[r];
             o_acc[sub][db] =
-                __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf[db], o_acc[sub][db], 0, 0, 0);
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf0, vf[0][db], o_acc[sub][db], 0, 0, 0);
+            o_acc[sub][db] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf1, vf[1][db], o_acc[sub][db], 0, 0, 0);
           }
         }
       }
       if (has_next) {
-        *(bf16x8*)(&k_lds[buf ^ 1][st_row][st_col]) = knext;
-        *(bf16x8*)(&v_tr[buf ^ 1][st_vt]) = vnext;
+        *(bf16x8*)(&k_lds[buf ^ 1][st_row][st_col]) = knext0;
+        *(bf16x8*)(&k_lds[buf ^ 1][st_row + 32][st_col]) = knext1;
+        *(bf16x8*)(&v_tr[buf ^ 1][st_vt]) = vnext0;
+        *(bf16x8*)(&v_tr[buf ^ 1][2048 + st_vt]) = vnext1;
       }
       __syncthreads(); // readers of buf done AND buf^1 writes visible
     }
@@ -280,7 +307,7 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor
   TORCH_CHECK(k.scalar_type() == at::kBFloat16 && v.scalar_type() == at::kBFloat16,
               "k/v must be bf16");
   const int B = q.size(0), H = q.size(1), N = q.size(2);
-  TORCH_CHECK(N % kKT == 0, "N must be a multiple of 32");
+  TORCH_CHECK(N % kKTF == 0, "N must be a multiple of 64");
   TORCH_CHECK(o.is_contiguous() && o.sizes() == q.sizes(), "o must be contiguous [B,H,N,D]");
   TORCH_CHECK(lse.scalar_type() == at::kFloat && lse.numel() >= (int64_t)B * H * N,
               "lse must be fp32[B*H*N]");
