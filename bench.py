@@ -133,8 +133,12 @@ class BenchStage(TrainValStage):
             self.pipeline.register_optimizer('opt', torch.optim.Adam(wrapped.parameters(), lr=1e-3))
 
         # synthetic data, resident on device: a pool of distinct batches
+        # (capped at ~16 GB so huge-batch configs leave HBM for activations)
         g = torch.Generator(device='cpu').manual_seed(4242)
-        self.n_pool = 16
+        batch_bytes = 4
+        for d in self.batch_shape:
+            batch_bytes *= d
+        self.n_pool = max(2, min(16, (16 << 30) // batch_bytes))
         if args.model == 'gpt2':
             vocab = self.pipeline.models['net'].module.cfg.vocab_size if hasattr(
                 self.pipeline.models['net'], 'module'

@@ -1,82 +1,90 @@
 """Low-level API example: manual epoch loop on the base Stage class.
 
-Mirror of the reference example (reference examples/barebone_mnist.py)
-on synthetic data, using the flat-replica fast path and fused Adam.
+The capability mirror of reference examples/barebone_mnist.py, but on
+the MI355X fast path: the model is a ``FlatReplica`` (all parameters in
+one flat HBM buffer), gradients sync with a single RCCL all-reduce, and
+the optimizer is the fused single-kernel ``FlatAdam``. Only ``run_epoch``
+is implemented by hand — everything TrainValStage would otherwise do
+(loops, loss tracking, prefixing) is spelled out here.
+
+Run:  python examples/barebone_mnist.py
+      torchrun --standalone --nproc-per-node 8 examples/barebone_mnist.py
 """
 
 import sys
+from pathlib import Path
 
-sys.path.insert(0, './')
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
-from torch import nn
+import torch.nn.functional as F
 from torch.utils.data import DataLoader
+from torch.utils.data.distributed import DistributedSampler
 
 from dmlcloud_amd import Stage, TrainingPipeline
 from dmlcloud_amd.models import SyntheticMnist, mnist_cnn
 from dmlcloud_amd.parallel import FlatAdam, FlatReplica, init_process_group_auto
 
 
-class MNISTStage(Stage):
+class BareboneMnistStage(Stage):
     def pre_stage(self):
-        train_dataset = SyntheticMnist(n=8192)
-        self.train_sampler = torch.utils.data.distributed.DistributedSampler(train_dataset)
-        self.train_loader = DataLoader(train_dataset, batch_size=32, sampler=self.train_sampler)
+        train_set = SyntheticMnist(n=8192)
+        self.train_sampler = DistributedSampler(train_set)
+        self.train_loader = DataLoader(train_set, batch_size=32, sampler=self.train_sampler)
 
-        val_dataset = SyntheticMnist(n=1024, seed=1)
-        val_sampler = torch.utils.data.distributed.DistributedSampler(val_dataset, shuffle=False)
-        self.val_loader = DataLoader(val_dataset, batch_size=32, sampler=val_sampler)
+        val_set = SyntheticMnist(n=1024, seed=1)
+        self.val_loader = DataLoader(val_set, batch_size=32, sampler=DistributedSampler(val_set, shuffle=False))
 
-        self.replica = FlatReplica(mnist_cnn().to(self.pipeline.device))
+        # one flat fp32 parameter buffer; broadcast from rank 0 at init
+        self.replica = FlatReplica(mnist_cnn().to(self.device))
         self.optimizer = FlatAdam(self.replica, lr=1e-3)
-        self.loss = nn.CrossEntropyLoss()
 
     def run_epoch(self):
-        self._train_epoch()
-        self._val_epoch()
-
-    def _train_epoch(self):
-        self.replica.module.train()
-        self.metric_prefix = 'train'
         self.train_sampler.set_epoch(self.current_epoch)
 
-        for img, target in self.train_loader:
-            img, target = img.to(self.pipeline.device), target.to(self.pipeline.device)
-            self.replica.zero_grad()
-            output = self.replica(img)
-            loss = self.loss(output, target)
-            loss.backward()
-            self.replica.grad_sync()
-            self.optimizer.step()
-            self._log_metrics(output, target, loss)
+        self.metric_prefix = 'train'
+        self.replica.module.train()
+        for batch in self.train_loader:
+            self._observe(*self._forward(batch), train=True)
 
-    @torch.no_grad()
-    def _val_epoch(self):
-        self.replica.module.eval()
         self.metric_prefix = 'val'
-        for img, target in self.val_loader:
-            img, target = img.to(self.pipeline.device), target.to(self.pipeline.device)
-            output = self.replica(img)
-            loss = self.loss(output, target)
-            self._log_metrics(output, target, loss)
+        self.replica.module.eval()
+        with torch.no_grad():
+            for batch in self.val_loader:
+                self._observe(*self._forward(batch), train=False)
 
-    def _log_metrics(self, output, target, loss):
+    def _forward(self, batch):
+        img, target = (t.to(self.device) for t in batch)
+        logits = self.replica(img)
+        return logits, target
+
+    def _observe(self, logits, target, train):
+        loss = F.cross_entropy(logits, target)
+        if train:
+            self.replica.zero_grad()
+            loss.backward()
+            self.replica.grad_sync()  # ONE RCCL all-reduce of the flat grads
+            self.optimizer.step()  # fused gfx950 Adam kernel
         self.track_reduce('loss', loss)
-        self.track_reduce('accuracy', (output.argmax(1) == target).float().mean())
+        self.track_reduce('accuracy', (logits.argmax(dim=1) == target).float().mean())
 
     def table_columns(self):
+        extra = [
+            ('[Train] Loss', 'train/loss'),
+            ('[Val] Loss', 'val/loss'),
+            ('[Train] Acc.', 'train/accuracy'),
+            ('[Val] Acc.', 'val/accuracy'),
+        ]
         columns = super().table_columns()
-        columns.insert(1, {'name': '[Train] Loss', 'metric': 'train/loss'})
-        columns.insert(2, {'name': '[Val] Loss', 'metric': 'val/loss'})
-        columns.insert(3, {'name': '[Train] Acc.', 'metric': 'train/accuracy'})
-        columns.insert(4, {'name': '[Val] Acc.', 'metric': 'val/accuracy'})
+        for offset, (title, metric) in enumerate(extra):
+            columns.insert(1 + offset, {'name': title, 'metric': metric})
         return columns
 
 
 def main():
     init_process_group_auto()
     pipeline = TrainingPipeline()
-    pipeline.append_stage(MNISTStage(), max_epochs=3)
+    pipeline.append_stage(BareboneMnistStage(), max_epochs=3)
     pipeline.run()
 
 
