@@ -60,6 +60,13 @@ def parse_args():
     p.add_argument('--no-channels-last', action='store_true', help='disable NHWC memory format (resnet50)')
     p.add_argument('--seq-len', type=int, default=1024, help='gpt2 sequence length')
     p.add_argument(
+        '--ckpt-layers',
+        default='',
+        help='comma-separated resnet50 stages (e.g. layer1,layer2) to run with '
+        'activation recompute — required to fit the bs=8192 reducer-stress '
+        'config in 288 GB HBM (activations alone are ~277 GB without it)',
+    )
+    p.add_argument(
         '--metric-stress',
         type=int,
         default=0,
@@ -103,6 +110,15 @@ class BenchStage(TrainValStage):
             if args.channels_last:
                 model = model.to(memory_format=torch.channels_last)
                 torch.backends.cudnn.benchmark = True
+            if args.ckpt_layers:
+                from torch.utils.checkpoint import checkpoint
+
+                def _recompute(stage_forward):
+                    return lambda x: checkpoint(stage_forward, x, use_reentrant=False)
+
+                for stage_name in args.ckpt_layers.split(','):
+                    stage = getattr(model, stage_name)
+                    stage.forward = _recompute(stage.forward)
             self.batch_shape = (args.batch_size, 3, 224, 224)
             self.dtype = 'bf16'
         else:  # gpt2
@@ -336,6 +352,7 @@ def main():
                 'hipgraph': bool(use_graph and getattr(graphed, 'captured', False)) if use_graph else False,
                 **({'seq_len': args.seq_len} if args.model == 'gpt2' else {}),
                 **({'metric_stress': args.metric_stress} if args.metric_stress else {}),
+                **({'ckpt_layers': args.ckpt_layers} if args.ckpt_layers else {}),
             },
         }
         print(json.dumps(result))
