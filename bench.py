@@ -243,6 +243,18 @@ def main():
         os.environ.setdefault('PYTORCH_TUNABLEOP_ENABLED', '1')
         os.environ.setdefault('PYTORCH_TUNABLEOP_TUNING', '1')
         os.environ.setdefault('PYTORCH_TUNABLEOP_FILENAME', f'gpurun_out/tunableop_{os.environ.get("RANK", 0)}.csv')
+    else:
+        # pre-tuned GEMM algorithm selections shipped per model config
+        # (profiles/tunableop_<model>_gfx950.csv, produced once with
+        # --tunableop and committed): load read-only — no tuning cost,
+        # every box starts with the searched hipBLASLt algorithms
+        pretuned = os.path.join(
+            os.path.dirname(os.path.abspath(__file__)), 'profiles', f'tunableop_{args.model}_gfx950.csv'
+        )
+        if torch.cuda.is_available() and os.path.exists(pretuned):
+            torch.cuda.tunable.enable(True)
+            torch.cuda.tunable.tuning_enable(False)
+            torch.cuda.tunable.read_file(pretuned)
 
     if 'MASTER_ADDR' not in os.environ and args.gpus == 1:
         pass  # dummy single-process group below
