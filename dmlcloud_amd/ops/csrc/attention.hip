@@ -74,12 +74,15 @@ __device__ __forceinline__ const __hip_bfloat16* tslice(const __hip_bfloat16* t,
   return t + (int64_t)(bh / H) * st.b + (int64_t)(bh % H) * st.h;
 }
 
-__global__ void __launch_bounds__(kWavesPerBlock * kWave)
-    __attribute__((amdgpu_waves_per_eu(3))) attn_fwd_kernel(
-        const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
-        const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
-        float* __restrict__ lse, int BH, int H, int N, float scale, bool causal, Strides sq,
-        Strides sk, Strides sv) {
+// Occupancy note: 204 VGPRs + 32 AGPRs -> 2 waves/SIMD. Forcing 3
+// waves/SIMD via amdgpu_waves_per_eu(3) spills 164 B/lane and measured
+// SLOWER (552 vs 508 us causal at the GPT-2 shape) — the spill traffic
+// in the inner loop outweighs the extra latency hiding. Keep 2 waves.
+__global__ void __launch_bounds__(kWavesPerBlock * kWave) attn_fwd_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
+    float* __restrict__ lse, int BH, int H, int N, float scale, bool causal, Strides sq,
+    Strides sk, Strides sv) {
   __shared__ __hip_bfloat16 p_lds_all[kWavesPerBlock][kQT][kPStrideF];
   __shared__ __hip_bfloat16 k_lds[2][kKTF][kKVStride];
   // V lives in TILED images read by ds_read_b64_tr_b16 (empirically
