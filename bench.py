@@ -109,7 +109,11 @@ class BenchStage(TrainValStage):
             args.channels_last = not args.no_channels_last and device.type == 'cuda'
             if args.channels_last:
                 model = model.to(memory_format=torch.channels_last)
-                torch.backends.cudnn.benchmark = True
+                # benchmark mode = exhaustive MIOpen find per conv shape:
+                # worth minutes of startup at b<=512, but at the b8192
+                # stress config the search itself takes >10 min — use
+                # MIOpen immediate mode there instead
+                torch.backends.cudnn.benchmark = args.batch_size < 2048
             if args.ckpt_layers:
                 from torch.utils.checkpoint import checkpoint
 
