@@ -292,6 +292,34 @@ class TestPrefetchBatch:
         assert list(ds) == [[0, 1, 2, 3], [4, 5, 6, 7]]
         assert len(ds) == 2
 
+    def test_prefetch_propagates_producer_error(self):
+        def exploding():
+            yield 1
+            yield 2
+            raise RuntimeError('boom in producer')
+
+        ds = PrefetchDataset(exploding(), 2)
+        it = iter(ds)
+        assert next(it) == 1
+        assert next(it) == 2
+        with pytest.raises(RuntimeError, match='boom in producer'):
+            next(it)
+
+    def test_prefetch_stays_ahead(self):
+        produced = []
+
+        def tracking():
+            for i in range(6):
+                produced.append(i)
+                yield i
+
+        ds = PrefetchDataset(tracking(), 3)
+        it = iter(ds)
+        first = next(it)
+        assert first == 0
+        # the producer thread ran ahead of the consumer
+        assert len(produced) >= 2
+
 
 class TestInterleave:
     def test_content(self):

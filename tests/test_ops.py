@@ -161,6 +161,26 @@ class TestClip:
         torch.nn.utils.clip_grad_norm_([p], 0.7)
         torch.testing.assert_close(g1, p.grad, rtol=1e-4, atol=1e-6)
 
+    def test_norm_scale_emulates_world_averaging(self):
+        """Clipping a W-summed gradient with norm_scale=1/W == clipping
+        the averaged gradient then re-summing (the flat-path contract)."""
+        torch.manual_seed(1)
+        world = 4
+        g_avg = torch.randn(1000)
+        g_sum = g_avg * world
+
+        norm = ops.clip_grad_norm_(g_sum, 0.5, norm_scale=1.0 / world)
+        assert norm[0].item() == pytest.approx(g_avg.norm().item(), rel=1e-4)
+
+        p = torch.nn.Parameter(torch.zeros(1000))
+        p.grad = g_avg.clone()
+        torch.nn.utils.clip_grad_norm_([p], 0.5)
+        torch.testing.assert_close(g_sum / world, p.grad, rtol=1e-4, atol=1e-6)
+
+    def test_l2_norm_scale(self):
+        x = torch.tensor([3.0, 4.0])
+        assert ops.l2_norm(x, norm_scale=0.5).item() == pytest.approx(2.5)
+
 
 class TestChunkedCopy:
     def test_cpu_pack(self):

@@ -43,6 +43,12 @@ class TestEnumAction:
         with pytest.raises(TypeError):
             parser.add_argument('--y', type=int, action=EnumAction)
 
+    def test_nargs_list(self):
+        parser = argparse.ArgumentParser()
+        parser.add_argument('--colors', type=Color, action=EnumAction, nargs='+')
+        args = parser.parse_args(['--colors', 'red', 'blue'])
+        assert args.colors == [Color.RED, Color.BLUE]
+
 
 class TestTcp:
     def test_free_port_bindable(self):
@@ -74,6 +80,13 @@ class TestThirdparty:
         assert try_import('definitely_not_a_module_xyz') is None
         assert try_get_version('definitely_not_a_module_xyz') is None
 
+    def test_installed_versions(self):
+        from dmlcloud_amd.utils.thirdparty import installed_versions
+
+        versions = installed_versions()
+        assert 'torch' in versions and 'numpy' in versions
+        assert all(isinstance(v, str) for v in versions.values())
+
 
 class TestSeed:
     def test_seed_reproducible(self):
@@ -87,6 +100,17 @@ class TestSeed:
 class TestLogging:
     def test_devnull(self):
         DevNullIO().write('anything')
+
+    def test_io_redirector_context_and_idempotence(self, tmp_path):
+        log = tmp_path / 'log.txt'
+        red = IORedirector(log)
+        with red:
+            red.install()  # second install is a no-op
+            print('inside-context')
+        assert not red.active
+        assert 'inside-context' in log.read_text()
+        # uninstall after uninstall is also a no-op
+        red.uninstall()
 
     def test_io_redirector(self, tmp_path):
         log = tmp_path / 'log.txt'
