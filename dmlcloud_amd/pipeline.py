@@ -224,8 +224,14 @@ class TrainingPipeline:
         for name, sd in state['models'].items():
             if name in self.models:
                 model = self.models[name]
-                module = model.module if hasattr(model, 'module') else model
-                module.load_state_dict(sd)
+                if isinstance(model, FlatReplica):
+                    # the replica's own load refreshes the fp32 master
+                    # (a later optimizer load restores the exact one)
+                    model.load_state_dict(sd)
+                elif hasattr(model, 'module'):
+                    model.module.load_state_dict(sd)
+                else:
+                    model.load_state_dict(sd)
         for name, sd in state['optimizers'].items():
             if name in self.optimizers:
                 self.optimizers[name].load_state_dict(sd)

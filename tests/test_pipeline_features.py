@@ -118,6 +118,49 @@ class TestFlatCheckpointRoundtrip:
         torch.testing.assert_close(opt2.exp_avg, opt.exp_avg)
         assert opt2.step_t.item() == 2
 
+    def test_pipeline_load_refreshes_flat_master(self, torch_distributed, tmp_path):
+        """pipeline.load_checkpoint must route through FlatReplica's own
+        load_state_dict so the fp32 master follows the loaded weights even
+        when no optimizer state is restored for that model."""
+        from dmlcloud_amd.parallel import FlatAdam
+
+        class FlatStage(Stage_):
+            def pre_stage(self):
+                torch.manual_seed(self.pipeline.config.get('seed', 0))
+                model = torch.nn.Linear(10, 10)
+                self.pipeline.register_model('m', model, ddp_impl='flat', flat_dtype=torch.bfloat16)
+                replica = self.pipeline.models['m']
+                self.pipeline.register_optimizer('adam', FlatAdam(replica, lr=1e-3))
+                self.pipeline.register_dataset('train', torch.utils.data.DataLoader(DS(), batch_size=4))
+                self.pipeline.register_dataset('val', torch.utils.data.DataLoader(DS(), batch_size=4))
+                self.loss = torch.nn.CrossEntropyLoss()
+
+            def step(self, batch):
+                x, y = batch
+                return self.loss(self.pipeline.models['m'](x.to(torch.bfloat16)).float(), y)
+
+        pipeline = TrainingPipeline(config={'seed': 0}, name='flatckpt')
+        pipeline.enable_checkpointing(str(tmp_path), resume=False)
+        pipeline.append_stage(FlatStage(), max_epochs=1)
+        pipeline.run()
+        pipeline.save_checkpoint()
+        trained_param = pipeline.models['m'].flat_param.clone()
+
+        # fresh pipeline with DIFFERENT init; drop the optimizer state so
+        # only the model load can refresh the master
+        pipeline2 = TrainingPipeline(config={'seed': 123}, name='flatckpt')
+        pipeline2.enable_checkpointing(str(tmp_path / 'other'), resume=False)
+        stage2 = FlatStage()
+        pipeline2.append_stage(stage2, max_epochs=1)
+        pipeline2.run()
+        pipeline2.checkpoint_dir = pipeline.checkpoint_dir
+        pipeline2.optimizers.clear()
+        pipeline2.load_checkpoint()
+
+        replica2 = pipeline2.models['m']
+        torch.testing.assert_close(replica2.flat_param, trained_param)
+        torch.testing.assert_close(replica2.flat_master, replica2.flat_param.to(torch.float32))
+
 
 class TestGraphedStepCpu:
     def test_eager_fallback(self):
