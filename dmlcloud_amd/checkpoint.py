@@ -246,13 +246,17 @@ def save_tensor_state(obj, path: Union[str, Path]):
         f.write(len(header).to_bytes(8, 'little'))
         f.write(header)
         if total:
-            f.write(host_buf.numpy().tobytes())
+            # memoryview over the pinned host buffer: no duplicate of the
+            # (potentially multi-GB) payload before the write syscall
+            f.write(memoryview(host_buf.numpy()))
     tmp.replace(path)  # atomic publish
 
 
 def load_tensor_state(path: Union[str, Path], device: Union[str, torch.device, None] = None):
     """Load a .dmlt file. With a CUDA device, the payload is uploaded with
     ONE H2D copy and tensors are zero-copy views of the flat buffer."""
+    import numpy as np
+
     path = Path(path)
     with open(path, 'rb') as f:
         magic = f.read(8)
@@ -260,16 +264,16 @@ def load_tensor_state(path: Union[str, Path], device: Union[str, torch.device, N
             raise ValueError(f'{path} is not a .dmlt checkpoint (bad magic {magic!r})')
         header_len = int.from_bytes(f.read(8), 'little')
         header = pickle.loads(f.read(header_len))
-        payload = f.read(header['total'])
+        total = header['total']
+        # read straight into the destination buffer: one file read, no
+        # intermediate bytes object for the payload
+        host_np = np.empty(total, dtype=np.uint8)
+        got = f.readinto(memoryview(host_np))
+        if got != total:
+            raise ValueError(f'{path}: truncated payload ({got} of {total} bytes)')
 
     metas = header['metas']
-    total = header['total']
-    if total:
-        import numpy as np
-
-        host_flat = torch.from_numpy(np.frombuffer(payload, dtype=np.uint8).copy())
-    else:
-        host_flat = torch.empty(0, dtype=torch.uint8)
+    host_flat = torch.from_numpy(host_np) if total else torch.empty(0, dtype=torch.uint8)
 
     device = torch.device(device) if device is not None else None
     if device is not None and device.type == 'cuda':
