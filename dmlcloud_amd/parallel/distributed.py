@@ -8,15 +8,21 @@ root-rank utilities, object collectives), redesigned for the ROCm stack:
   backend string ``cpu:gloo,cuda:nccl`` gives RCCL-over-xGMI for device
   tensors and gloo for CPU control-plane tensors.
 - Fixes the reference's env:// gap (reference util/distributed.py:237-238
-  leaves _WorkerInfo unpopulated so local_rank() is None under torchrun):
-  here the env path reads RANK/WORLD_SIZE/LOCAL_RANK/LOCAL_WORLD_SIZE.
+  leaves the worker topology unpopulated so local_rank() is None under
+  torchrun): here the env path reads RANK/WORLD_SIZE/LOCAL_RANK/
+  LOCAL_WORLD_SIZE.
 - ``HSA_ENABLE_IPC_MODE_LEGACY=0`` is required on this host driver for
-  RCCL dmabuf IPC; init_process_group_auto asserts it is not overridden.
+  RCCL dmabuf IPC.
+
+Internal layout: worker topology lives in one ``_Topology`` record that
+each ``init_process_group_*`` entry point fills before the common
+``_start_group`` helper brings up torch.distributed.
 """
 
 import datetime
 import os
 from contextlib import contextmanager
+from dataclasses import dataclass, field, fields
 from typing import Optional
 
 import torch
@@ -27,30 +33,70 @@ from ..utils.tcp import find_free_port, get_local_ips
 DEFAULT_PORT = int(os.environ.get('DMLCLOUD_PORT', 41312))  # 41312 = "dml"
 
 
-class _WorkerInfo:
-    INIT_METHOD = None
-    RANK = None
-    WORLD_SIZE = None
-    LOCAL_RANK = None
-    LOCAL_WORLD_SIZE = None
-    NODE_ID = None
+@dataclass
+class _Topology:
+    """How this process fits into the job (filled at init, read by the
+    rank()/local_rank()/... accessors; all None before init)."""
+
+    method: Optional[str] = None
+    rank: Optional[int] = None
+    world: Optional[int] = None
+    local_rank: Optional[int] = None
+    local_world: Optional[int] = None
+    node: Optional[int] = None
+
+    def reset(self):
+        for f in fields(self):
+            setattr(self, f.name, None)
+
+
+_topo = _Topology()
+
+
+# ------------------------------------------------------- launcher probes
+
+
+def has_environment() -> bool:
+    """torchrun / torch.distributed.run sets the MASTER_* rendezvous vars."""
+    return 'MASTER_PORT' in os.environ
 
 
 def has_slurm() -> bool:
     return 'SLURM_PROCID' in os.environ
 
 
-def has_environment() -> bool:
-    return 'MASTER_PORT' in os.environ
-
-
 def has_mpi() -> bool:
     try:
-        from mpi4py import MPI  # noqa: F401
-
-        return True
+        import mpi4py  # noqa: F401
     except ImportError:
         return False
+    return True
+
+
+# ------------------------------------------------------ topology accessors
+
+
+def rank() -> Optional[int]:
+    return _topo.rank
+
+
+def world_size() -> Optional[int]:
+    return _topo.world
+
+
+def local_rank() -> Optional[int]:
+    return _topo.local_rank
+
+
+def local_world_size() -> Optional[int]:
+    return _topo.local_world
+
+
+def local_node() -> Optional[int]:
+    return _topo.node
+
+
+# ---------------------------------------------------------- root utilities
 
 
 def is_root() -> bool:
@@ -80,51 +126,17 @@ def root_first():
             dist.barrier()
     else:
         dist.barrier()
-        try:
-            yield
-        finally:
-            pass
-
-
-def mpi_local_comm():
-    try:
-        from mpi4py import MPI
-
-        comm = MPI.COMM_WORLD
-        return comm.Split_type(MPI.COMM_TYPE_SHARED, 0, MPI.INFO_NULL)
-    except ImportError:
-        return None
-
-
-def rank() -> Optional[int]:
-    return _WorkerInfo.RANK
-
-
-def world_size() -> Optional[int]:
-    return _WorkerInfo.WORLD_SIZE
-
-
-def local_rank() -> Optional[int]:
-    return _WorkerInfo.LOCAL_RANK
-
-
-def local_world_size() -> Optional[int]:
-    return _WorkerInfo.LOCAL_WORLD_SIZE
-
-
-def local_node() -> Optional[int]:
-    return _WorkerInfo.NODE_ID
+        yield
 
 
 def print_worker(msg, barrier: bool = True, flush: bool = True):
     """Rank-tagged debug printing, optionally fenced by barriers for ordering."""
     if barrier:
         dist.barrier()
-    s = f'Worker {rank()}'
+    tag = f'Worker {rank()}'
     if local_node() is not None:
-        s += f'({local_node()}.{local_rank()})'
-    s += f': {msg}'
-    print(s, flush=flush)
+        tag = f'{tag}({local_node()}.{local_rank()})'
+    print(f'{tag}: {msg}', flush=flush)
     if barrier:
         dist.barrier()
 
@@ -134,25 +146,28 @@ def print_root(msg, flush: bool = True):
     print(msg, flush=flush)
 
 
+# -------------------------------------------------------- object collectives
+
+
 def all_gather_object(obj, group=None):
-    outlist = [None for _ in range(dist.get_world_size(group))]
-    dist.all_gather_object(outlist, obj, group=group)
-    return outlist
+    gathered = [None] * dist.get_world_size(group)
+    dist.all_gather_object(gathered, obj, group=group)
+    return gathered
 
 
 def gather_object(obj, dst: int = 0, group=None):
-    if dist.get_rank() == dst:
-        outlist = [None for _ in range(dist.get_world_size(group))]
-    else:
-        outlist = None
-    dist.gather_object(obj, outlist, dst=dst, group=group)
-    return outlist
+    gathered = [None] * dist.get_world_size(group) if dist.get_rank() == dst else None
+    dist.gather_object(obj, gathered, dst=dst, group=group)
+    return gathered
 
 
 def broadcast_object(obj, src: int = 0, group=None):
-    objlist = [obj]
-    dist.broadcast_object_list(objlist, src=src, group=group)
-    return objlist[0]
+    box = [obj]
+    dist.broadcast_object_list(box, src=src, group=group)
+    return box[0]
+
+
+# ----------------------------------------------------------- initialization
 
 
 def _default_backend() -> str:
@@ -161,22 +176,22 @@ def _default_backend() -> str:
     return 'gloo'
 
 
+def _start_group(kwargs, **init_args):
+    """Common torch.distributed bring-up: resolve the backend, then
+    init_process_group with whatever rendezvous the caller prepared."""
+    kwargs.setdefault('backend', _default_backend())
+    dist.init_process_group(**init_args, **kwargs)
+
+
 def init_process_group_dummy(**kwargs):
     """World-size-1 process group over a HashStore.
 
     Lets every collective code path execute for real in tests and
     single-GPU runs without a rendezvous server.
     """
-    _WorkerInfo.INIT_METHOD = 'dummy'
-    _WorkerInfo.RANK = 0
-    _WorkerInfo.WORLD_SIZE = 1
-    _WorkerInfo.LOCAL_RANK = 0
-    _WorkerInfo.LOCAL_WORLD_SIZE = 1
-    _WorkerInfo.NODE_ID = 0
-
-    backend = kwargs.pop('backend', None) or _default_backend()
-    store = dist.HashStore()
-    dist.init_process_group(store=store, rank=0, world_size=1, backend=backend, **kwargs)
+    _topo.method, _topo.rank, _topo.world = 'dummy', 0, 1
+    _topo.local_rank, _topo.local_world, _topo.node = 0, 1, 0
+    _start_group(kwargs, store=dist.HashStore(), rank=0, world_size=1)
 
 
 def init_process_group_env(**kwargs):
@@ -186,43 +201,49 @@ def init_process_group_env(**kwargs):
     torchrun env vars so local_rank() works and device selection can map
     rank -> GPU.
     """
-    _WorkerInfo.INIT_METHOD = 'env'
-    _WorkerInfo.RANK = int(os.environ['RANK']) if 'RANK' in os.environ else None
-    _WorkerInfo.WORLD_SIZE = int(os.environ['WORLD_SIZE']) if 'WORLD_SIZE' in os.environ else None
-    if 'LOCAL_RANK' in os.environ:
-        _WorkerInfo.LOCAL_RANK = int(os.environ['LOCAL_RANK'])
-    if 'LOCAL_WORLD_SIZE' in os.environ:
-        _WorkerInfo.LOCAL_WORLD_SIZE = int(os.environ['LOCAL_WORLD_SIZE'])
-    if 'GROUP_RANK' in os.environ:
-        _WorkerInfo.NODE_ID = int(os.environ['GROUP_RANK'])
 
-    kwargs.setdefault('backend', _default_backend())
-    dist.init_process_group(init_method='env://', **kwargs)
-    if _WorkerInfo.RANK is None:
-        _WorkerInfo.RANK = dist.get_rank()
-    if _WorkerInfo.WORLD_SIZE is None:
-        _WorkerInfo.WORLD_SIZE = dist.get_world_size()
+    def env_int(name):
+        return int(os.environ[name]) if name in os.environ else None
+
+    _topo.method = 'env'
+    _topo.rank = env_int('RANK')
+    _topo.world = env_int('WORLD_SIZE')
+    _topo.local_rank = env_int('LOCAL_RANK')
+    _topo.local_world = env_int('LOCAL_WORLD_SIZE')
+    _topo.node = env_int('GROUP_RANK')
+
+    _start_group(kwargs, init_method='env://')
+    if _topo.rank is None:
+        _topo.rank = dist.get_rank()
+    if _topo.world is None:
+        _topo.world = dist.get_world_size()
 
 
 def init_process_group_slurm(port: int = DEFAULT_PORT, **kwargs):
     """SLURM srun rendezvous over tcp://SLURM_SRUN_COMM_HOST."""
-    _WorkerInfo.INIT_METHOD = 'slurm'
-    _WorkerInfo.RANK = int(os.environ['SLURM_PROCID'])
-    _WorkerInfo.WORLD_SIZE = int(os.environ['SLURM_NTASKS'])
-    _WorkerInfo.LOCAL_RANK = int(os.environ['SLURM_LOCALID'])
+    env = os.environ
+    _topo.method = 'slurm'
+    _topo.rank = int(env['SLURM_PROCID'])
+    _topo.world = int(env['SLURM_NTASKS'])
+    _topo.local_rank = int(env['SLURM_LOCALID'])
     # SLURM_STEP_TASKS_PER_NODE can be a list like "8(x2)"; take the first count
-    tasks_per_node = os.environ['SLURM_STEP_TASKS_PER_NODE'].split('(')[0].split(',')[0]
-    _WorkerInfo.LOCAL_WORLD_SIZE = int(tasks_per_node)
-    _WorkerInfo.NODE_ID = int(os.environ['SLURM_NODEID'])
+    _topo.local_world = int(env['SLURM_STEP_TASKS_PER_NODE'].split('(')[0].split(',')[0])
+    _topo.node = int(env['SLURM_NODEID'])
 
-    ip = os.environ['SLURM_SRUN_COMM_HOST']
-    kwargs.setdefault('backend', _default_backend())
-    dist.init_process_group(
-        init_method=f'tcp://{ip}:{port}',
-        world_size=_WorkerInfo.WORLD_SIZE,
-        rank=_WorkerInfo.RANK,
-        **kwargs,
+    _start_group(
+        kwargs,
+        init_method=f'tcp://{env["SLURM_SRUN_COMM_HOST"]}:{port}',
+        world_size=_topo.world,
+        rank=_topo.rank,
     )
+
+
+def mpi_local_comm():
+    try:
+        from mpi4py import MPI
+    except ImportError:
+        return None
+    return MPI.COMM_WORLD.Split_type(MPI.COMM_TYPE_SHARED, 0, MPI.INFO_NULL)
 
 
 def init_process_group_MPI(ip_idx: int = 0, port: Optional[int] = DEFAULT_PORT, **kwargs):
@@ -232,28 +253,26 @@ def init_process_group_MPI(ip_idx: int = 0, port: Optional[int] = DEFAULT_PORT, 
     from mpi4py import MPI
 
     comm = MPI.COMM_WORLD
-    local_comm = mpi_local_comm()
+    node_comm = mpi_local_comm()
 
-    _WorkerInfo.INIT_METHOD = 'mpi'
-    _WorkerInfo.RANK = comm.Get_rank()
-    _WorkerInfo.WORLD_SIZE = comm.Get_size()
-    _WorkerInfo.LOCAL_RANK = local_comm.Get_rank()
-    _WorkerInfo.LOCAL_WORLD_SIZE = local_comm.Get_size()
+    _topo.method = 'mpi'
+    _topo.rank = comm.Get_rank()
+    _topo.world = comm.Get_size()
+    _topo.local_rank = node_comm.Get_rank()
+    _topo.local_world = node_comm.Get_size()
 
     if port is None:
         port = find_free_port()
-
-    ip = get_local_ips()[ip_idx] if _WorkerInfo.RANK == 0 else None
-    ip = comm.bcast(ip, root=0)
+    address = get_local_ips()[ip_idx] if _topo.rank == 0 else None
+    address = comm.bcast(address, root=0)
     port = comm.bcast(port, root=0)
     comm.Barrier()
 
-    kwargs.setdefault('backend', _default_backend())
-    dist.init_process_group(
-        init_method=f'tcp://{ip}:{port}',
-        world_size=_WorkerInfo.WORLD_SIZE,
-        rank=_WorkerInfo.RANK,
-        **kwargs,
+    _start_group(
+        kwargs,
+        init_method=f'tcp://{address}:{port}',
+        world_size=_topo.world,
+        rank=_topo.rank,
     )
 
 
@@ -274,13 +293,11 @@ def init_process_group_auto(verbose: bool = True, **kwargs):
 
 def deinitialize_torch_distributed():
     """Tear down the process group and reset worker topology."""
-    _WorkerInfo.INIT_METHOD = None
-    _WorkerInfo.RANK = None
-    _WorkerInfo.WORLD_SIZE = None
-    _WorkerInfo.LOCAL_RANK = None
-    _WorkerInfo.LOCAL_WORLD_SIZE = None
-    _WorkerInfo.NODE_ID = None
+    _topo.reset()
     dist.destroy_process_group()
+
+
+# ----------------------------------------------------------------- barriers
 
 
 def new_gloo_group():

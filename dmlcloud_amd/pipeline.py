@@ -104,8 +104,7 @@ class TrainingPipeline:
         grad_sync, fused optimizers, hipGraph-friendly); 'none'/use_ddp=False
         leaves the module unwrapped (parameters broadcast from rank 0).
         """
-        if name in self.models:
-            raise ValueError(f'Model with name {name} already exists')
+        self._claim(self.models, 'Model', name)
 
         if use_ddp and ddp_impl == 'torch':
             model = wrap_ddp(model, self.device, sync_bn=sync_bn, bucket_cap_mb=bucket_cap_mb)
@@ -135,33 +134,35 @@ class TrainingPipeline:
         if verbose:
             module = model.module if hasattr(model, 'module') else model
             n_params = sum(p.numel() for p in module.parameters())
-            msg = f'Model "{name}":\n'
-            msg += f'    - Parameters: {n_params / 1e6:.1f} M\n'
-            msg += f'    - DDP: {use_ddp} ({ddp_impl})\n'
-            msg += f'    - {model}'
-            self.logger.info(msg)
+            self.logger.info(
+                '\n'.join(
+                    [
+                        f'Model "{name}":',
+                        f'    - Parameters: {n_params / 1e6:.1f} M',
+                        f'    - DDP: {use_ddp} ({ddp_impl})',
+                        f'    - {model}',
+                    ]
+                )
+            )
 
     def register_optimizer(self, name: str, optimizer, scheduler=None):
-        if name in self.optimizers:
-            raise ValueError(f'Optimizer with name {name} already exists')
+        self._claim(self.optimizers, 'Optimizer', name)
         self.optimizers[name] = optimizer
         if scheduler is not None:
             self.schedulers[name] = scheduler
 
     def register_dataset(self, name: str, dataset: Union[DataLoader, Dataset, Sequence], verbose: bool = True):
-        if name in self.datasets:
-            raise ValueError(f'Dataset with name {name} already exists')
+        self._claim(self.datasets, 'Dataset', name)
         self.datasets[name] = dataset
         if verbose:
-            msg = f'Dataset "{name}":\n'
             try:
-                length = len(dataset)
-                msg += f'    - Batches (Total): ~{length * dist.get_world_size()}\n'
-                msg += f'    - Batches (/Worker): {length}\n'
-            except TypeError:
-                msg += '    - Batches (Total): N/A\n'
-                msg += '    - Batches (/Worker): N/A\n'
-            self.logger.info(msg)
+                per_worker = str(len(dataset))
+                total = f'~{len(dataset) * dist.get_world_size()}'
+            except TypeError:  # length-less iterables
+                per_worker = total = 'N/A'
+            self.logger.info(
+                f'Dataset "{name}":\n    - Batches (Total): {total}\n    - Batches (/Worker): {per_worker}\n'
+            )
 
     def append_stage(self, stage: Stage, max_epochs: Optional[int] = None, name: Optional[str] = None):
         if not isinstance(stage, Stage):
@@ -170,6 +171,11 @@ class TrainingPipeline:
         stage.max_epochs = max_epochs
         stage.name = name
         self.stages.append(stage)
+
+    @staticmethod
+    def _claim(registry: dict, kind: str, name: str):
+        if name in registry:
+            raise ValueError(f'{kind} with name {name} already exists')
 
     # --------------------------------------------------------- checkpointing
 
@@ -321,6 +327,41 @@ class TrainingPipeline:
     def resume_run(self):
         pass
 
+    def _select_device(self) -> torch.device:
+        """Map this rank to its GPU (local_rank -> HIP device) or CPU."""
+        if not torch.cuda.is_available():
+            warnings.warn('No GPU available. Running on CPU.')
+            return torch.device('cpu')
+        rank_on_node = local_rank()
+        if rank_on_node is None:
+            warnings.warn(
+                'GPU is available but no local rank found. Make sure to set HIP_VISIBLE_DEVICES '
+                'manually for each rank.'
+            )
+            return torch.device('cuda')
+        torch.cuda.set_device(rank_on_node)
+        return torch.device('cuda', rank_on_node)
+
+    def _warm_communicator(self):
+        """First collective initializes the RCCL communicator lazily; do it
+        NOW so communicator setup (xGMI ring/tree discovery) never lands
+        inside the timed training path or a later hipGraph capture region."""
+        if self.device.type == 'cuda' and dist.get_world_size() > 1:
+            dist.all_reduce(torch.zeros(1, device=self.device))
+            torch.cuda.synchronize()
+
+    def _log_startup(self):
+        add_log_handlers(self.logger)
+        self.logger.info('\n' + experiment_header(self.name, self.checkpoint_dir, self.start_time))
+
+    def _log_diagnostics(self):
+        ranks = all_gather_object(str(self.device))
+        sections = [general_diagnostics(), '* DEVICES:']
+        sections += [f'    - [Rank {i}] {d}' for i, d in enumerate(ranks)]
+        sections.append('* CONFIG:')
+        sections += [f'    {line}' for line in self.config.to_yaml(resolve=True).splitlines()]
+        self.logger.info('\n'.join(sections))
+
     def _pre_run(self):
         if len(self.stages) == 0:
             raise ValueError('No stages defined. Use append_stage() to add stages to the pipeline.')
@@ -335,53 +376,22 @@ class TrainingPipeline:
         else:
             warnings.warn('Gloo backend not available. Barriers will not use custom timeouts.')
 
-        if torch.cuda.is_available():
-            if local_rank() is None:
-                warnings.warn(
-                    'GPU is available but no local rank found. Make sure to set HIP_VISIBLE_DEVICES '
-                    'manually for each rank.'
-                )
-                self.device = torch.device('cuda')
-            else:
-                self.device = torch.device('cuda', local_rank())
-                torch.cuda.set_device(local_rank())
-        else:
-            warnings.warn('No GPU available. Running on CPU.')
-            self.device = torch.device('cpu')
-
-        if self.device.type == 'cuda' and dist.get_world_size() > 1:
-            # First collective initializes the RCCL communicator lazily;
-            # do it NOW so communicator setup (xGMI ring/tree discovery)
-            # never lands inside the timed training path or a later
-            # hipGraph capture region.
-            dist.all_reduce(torch.zeros(1, device=self.device))
-            torch.cuda.synchronize()
+        self.device = self._select_device()
+        self._warm_communicator()
 
         # prevent checkpoint dir creation before all ranks searched for it
         self.barrier(timeout=10 * 60)
         if self.checkpointing_enabled:
             self._init_checkpointing()
-
         if self.wandb:
             self._wandb_initializer()
-
         self.barrier(timeout=10 * 60)
+
         self.start_time = datetime.now()
-
-        add_log_handlers(self.logger)
-        header = '\n' + experiment_header(self.name, self.checkpoint_dir, self.start_time)
-        self.logger.info(header)
-
+        self._log_startup()
         if self.resumed:
             self._resume_run()
-
-        diagnostics = general_diagnostics()
-        diagnostics += '\n* DEVICES:\n'
-        devices = all_gather_object(str(self.device))
-        diagnostics += '\n'.join(f'    - [Rank {i}] {device}' for i, device in enumerate(devices))
-        diagnostics += '\n* CONFIG:\n'
-        diagnostics += '\n'.join(f'    {line}' for line in self.config.to_yaml(resolve=True).splitlines())
-        self.logger.info(diagnostics)
+        self._log_diagnostics()
 
         self.pre_run()
 
