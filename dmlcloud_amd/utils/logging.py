@@ -29,7 +29,7 @@ import torch
 import torch.distributed as dist
 
 from .git import git_hash
-from .slurm import slurm_job_id, slurm_step_id
+from .slurm import slurm_summary
 from .thirdparty import installed_versions
 
 __all__ = [
@@ -230,16 +230,8 @@ def general_diagnostics() -> str:
     versions += sorted(installed_versions().items())
     report += _bullet_block('VERSIONS', versions)
 
-    if slurm_job_id() is not None:
-        slurm_vars = [('SLURM_JOB_ID', slurm_job_id()), ('SLURM_STEP_ID', slurm_step_id())]
-        for env_name in (
-            'SLURM_STEP_NODELIST',
-            'SLURM_TASKS_PER_NODE',
-            'SLURM_STEP_GPUS',
-            'SLURM_GPUS_ON_NODE',
-            'SLURM_CPUS_PER_TASK',
-        ):
-            slurm_vars.append((env_name, os.environ.get(env_name)))
-        report += _bullet_block('SLURM', slurm_vars)
+    slurm_facts = slurm_summary()
+    if slurm_facts:
+        report += _bullet_block('SLURM', slurm_facts.items())
 
     return report.rstrip('\n')
