@@ -96,6 +96,36 @@ class TestSeed:
         b = torch.randn(4)
         torch.testing.assert_close(a, b)
 
+    def test_seed_for_rank_diverges_per_rank(self):
+        from dmlcloud_amd.utils.seed import seed_for_rank
+
+        s0 = seed_for_rank(7, rank_=0)
+        a = torch.randn(4)
+        s1 = seed_for_rank(7, rank_=1)
+        b = torch.randn(4)
+        assert s0 != s1
+        assert not torch.equal(a, b)
+        # deterministic per (seed, rank)
+        assert seed_for_rank(7, rank_=0) == s0
+        torch.testing.assert_close(torch.randn(4), a)
+
+
+class TestSlurmSummary:
+    def test_off_slurm_empty(self, monkeypatch):
+        monkeypatch.delenv('SLURM_JOB_ID', raising=False)
+        from dmlcloud_amd.utils.slurm import slurm_summary
+
+        assert slurm_summary() == {}
+
+    def test_on_slurm_reports(self, monkeypatch):
+        monkeypatch.setenv('SLURM_JOB_ID', '42')
+        monkeypatch.setenv('SLURM_CPUS_PER_TASK', '16')
+        from dmlcloud_amd.utils.slurm import slurm_summary
+
+        facts = slurm_summary()
+        assert facts['SLURM_JOB_ID'] == '42'
+        assert facts['SLURM_CPUS_PER_TASK'] == '16'
+
 
 class TestLogging:
     def test_devnull(self):
