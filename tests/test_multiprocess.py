@@ -326,6 +326,19 @@ def _pipeline_two_ranks(rank, world_size, tmpdir):
     torch.testing.assert_close(gathered[0], gathered[1])
 
 
+def _checkpoint_path_agreement(rank, world_size, tmpdir):
+    """enable_checkpointing broadcasts the generated dir so every rank
+    agrees (collective C7 in SURVEY §2.5); only root creates it."""
+    from dmlcloud_amd import TrainingPipeline
+    from dmlcloud_amd.parallel import all_gather_object
+
+    pipeline = TrainingPipeline(name='agree')
+    pipeline.enable_checkpointing(tmpdir, resume=False)
+    paths = all_gather_object(str(pipeline.checkpoint_dir))
+    assert len(set(paths)) == 1, paths
+    assert not pipeline.resumed
+
+
 def _root_helpers(rank, world_size, tmpdir):
     from dmlcloud_amd.parallel import all_gather_object, broadcast_object, gather_object, is_root
 
@@ -355,6 +368,7 @@ def _root_helpers(rank, world_size, tmpdir):
         '_flat_clip_matches_ddp',
         '_flat_overlap_no_sync_accumulation',
         '_metric_name_order_divergence_raises',
+        '_checkpoint_path_agreement',
         '_pipeline_two_ranks',
         '_root_helpers',
     ],
