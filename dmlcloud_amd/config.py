@@ -23,7 +23,9 @@ from typing import Any, Dict, Optional, Union
 
 import yaml
 
-_INTERP_RE = re.compile(r'\$\{([a-zA-Z0-9_.]+)\}')
+# \w covers unicode identifiers too (any key YAML can express and Python
+# can attribute-access); dots path into nested configs
+_INTERP_RE = re.compile(r'\$\{([\w.]+)\}')
 
 
 class Config:

@@ -137,5 +137,44 @@ def test_config_yaml_roundtrip(data, tmp_path_factory):
     assert Config.load(path) == cfg
 
 
+@settings(max_examples=60, deadline=None)
+@given(
+    num_batches=st.integers(1, 4),
+    slice_size=st.integers(1, 6),
+    cols=st.integers(1, 5),
+    seed=st.integers(0, 100),
+)
+def test_interleave_is_content_preserving(num_batches, slice_size, cols, seed):
+    """Every input element appears exactly once in the interleaved output,
+    and batch shapes are preserved."""
+    import torch
+
+    from dmlcloud_amd.data import interleave_batches
+
+    torch.manual_seed(seed)
+    batch_rows = slice_size * num_batches
+    batches = [torch.randn(batch_rows, cols) for _ in range(num_batches)]
+    out = [b.clone() for b in interleave_batches(iter(batches), num_batches)]
+    assert len(out) == num_batches
+    assert all(o.shape == batches[0].shape for o in out)
+    source = torch.cat(batches).flatten().sort().values
+    produced = torch.cat(out).flatten().sort().values
+    torch.testing.assert_close(source, produced)
+
+
+@settings(max_examples=80, deadline=None)
+@given(
+    key=st.text(st.characters(whitelist_categories=('Ll',)), min_size=1, max_size=6),
+    value=st.one_of(st.integers(), st.floats(allow_nan=False, allow_infinity=False), st.booleans()),
+)
+def test_config_interpolation_resolves_to_referenced_value(key, value):
+    """Whole-string ${key} interpolation yields the referenced value with
+    its type preserved."""
+    cfg = Config.create({key: value, 'alias': '${' + key + '}'})
+    resolved = cfg.to_container(resolve=True)
+    assert resolved['alias'] == value
+    assert type(resolved['alias']) is type(value)
+
+
 if __name__ == '__main__':
     sys.exit(pytest.main([__file__]))
