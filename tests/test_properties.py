@@ -176,5 +176,69 @@ def test_config_interpolation_resolves_to_referenced_value(key, value):
     assert type(resolved['alias']) is type(value)
 
 
+def _state_strategy():
+    """Nested state objects shaped like real checkpoints: dicts/lists of
+    tensors (varied dtypes/shapes incl. 0-dim and empty) and scalars."""
+    import torch
+
+    def tensor_strategy():
+        dtypes = st.sampled_from([torch.float32, torch.float64, torch.bfloat16, torch.int64, torch.uint8, torch.bool])
+        shapes = st.sampled_from([(), (1,), (3,), (2, 3), (0,), (4, 1, 2)])
+        seeds = st.integers(0, 99)
+
+        def make(args):
+            dtype, shape, seed = args
+            g = torch.Generator().manual_seed(seed)
+            if dtype is torch.bool:
+                return torch.rand(shape, generator=g) > 0.5
+            if dtype in (torch.int64, torch.uint8):
+                return torch.randint(0, 100, shape, generator=g, dtype=dtype)
+            return torch.randn(shape, generator=g).to(dtype)
+
+        return st.tuples(dtypes, shapes, seeds).map(make)
+
+    leaves = st.one_of(tensor_strategy(), st.integers(), st.floats(allow_nan=False), st.text(max_size=5), st.none())
+    return st.recursive(
+        leaves,
+        lambda children: st.one_of(
+            st.dictionaries(st.text(max_size=4), children, max_size=3),
+            st.lists(children, max_size=3),
+        ),
+        max_leaves=8,
+    )
+
+
+@settings(max_examples=40, deadline=None)
+@given(state=_state_strategy(), seed=st.integers(0, 10))
+def test_dmlt_roundtrip_property(state, seed, tmp_path_factory):
+    """Any nested checkpoint-shaped object survives the .dmlt round-trip
+    with identical structure, dtypes, shapes and values."""
+    import torch
+
+    from dmlcloud_amd.checkpoint import load_tensor_state, save_tensor_state
+
+    path = tmp_path_factory.mktemp('dmlt') / 's.dmlt'
+    save_tensor_state(state, path)
+    loaded = load_tensor_state(path)
+
+    def check(a, b):
+        if isinstance(a, torch.Tensor):
+            assert isinstance(b, torch.Tensor)
+            assert a.dtype == b.dtype and tuple(a.shape) == tuple(b.shape)
+            torch.testing.assert_close(b, a, equal_nan=True)
+        elif isinstance(a, dict):
+            assert set(a) == set(b)
+            for k in a:
+                check(a[k], b[k])
+        elif isinstance(a, (list, tuple)):
+            assert len(a) == len(b)
+            for x, y in zip(a, b):
+                check(x, y)
+        else:
+            assert a == b or (a != a and b != b)
+
+    check(state, loaded)
+
+
 if __name__ == '__main__':
     sys.exit(pytest.main([__file__]))
