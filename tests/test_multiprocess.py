@@ -350,6 +350,24 @@ def _root_helpers(rank, world_size, tmpdir):
         assert gathered == [0, 1]
 
 
+def _root_first_ordering(rank, world_size, tmpdir):
+    """root_first: rank 0's body completes before any other rank's starts
+    (the dataset-download fence)."""
+    import os
+    import time
+
+    from dmlcloud_amd.parallel.distributed import root_first
+
+    marker = os.path.join(tmpdir, 'root_done')
+    with root_first():
+        if rank == 0:
+            time.sleep(0.2)  # make a race observable if the fence is broken
+            with open(marker, 'w') as f:
+                f.write('ok')
+        else:
+            assert os.path.exists(marker), 'non-root entered before root finished'
+
+
 # --------------------------------------------------------------------- tests
 
 
@@ -371,6 +389,7 @@ def _root_helpers(rank, world_size, tmpdir):
         '_checkpoint_path_agreement',
         '_pipeline_two_ranks',
         '_root_helpers',
+        '_root_first_ordering',
     ],
 )
 def test_multiprocess(payload, tmp_path):

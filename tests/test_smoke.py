@@ -56,6 +56,23 @@ class TestSmoke:
         pipeline.run()
         assert len(pipeline.tracker['train/loss']) == 3
 
+    def test_two_stages_share_tracker(self, torch_distributed):
+        """Multiple stages run in order on one tracker; stage epochs keep
+        appending to the same histories."""
+
+        class NamedStage(DummyStage):
+            pass
+
+        pipeline = TrainingPipeline()
+        pipeline.append_stage(DummyStage(), max_epochs=2, name='warmup')
+        pipeline.append_stage(NamedStage(), max_epochs=1, name='finetune')
+        pipeline.run()
+        # 2 + 1 epochs of train/loss landed in one shared history
+        assert len(pipeline.tracker['train/loss']) == 3
+        assert all(v is not None for v in pipeline.tracker['train/loss'])
+        assert pipeline.stages[0].stop_time is not None
+        assert pipeline.stages[1].current_epoch == 2  # ran its single epoch
+
     def test_no_stages_raises(self, torch_distributed):
         pipeline = TrainingPipeline()
         with pytest.raises(ValueError):
