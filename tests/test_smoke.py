@@ -57,15 +57,17 @@ class TestSmoke:
         assert len(pipeline.tracker['train/loss']) == 3
 
     def test_two_stages_share_tracker(self, torch_distributed):
-        """Multiple stages run in order on one tracker; stage epochs keep
-        appending to the same histories."""
+        """Multiple stages run in order on one tracker; the second stage
+        reuses the registries the first one filled (warmup -> finetune)."""
 
-        class NamedStage(DummyStage):
-            pass
+        class FinetuneStage(DummyStage):
+            def pre_stage(self):
+                # model/optimizer/datasets already registered by stage 1
+                self.loss = torch.nn.CrossEntropyLoss()
 
         pipeline = TrainingPipeline()
         pipeline.append_stage(DummyStage(), max_epochs=2, name='warmup')
-        pipeline.append_stage(NamedStage(), max_epochs=1, name='finetune')
+        pipeline.append_stage(FinetuneStage(), max_epochs=1, name='finetune')
         pipeline.run()
         # 2 + 1 epochs of train/loss landed in one shared history
         assert len(pipeline.tracker['train/loss']) == 3
