@@ -16,8 +16,26 @@ import logging
 from typing import Callable, Optional
 
 import torch
+import torch.distributed as dist
 
 logger = logging.getLogger('dmlcloud_amd')
+
+
+def _capture_unanimous(captured: bool) -> bool:
+    """Vote across ranks on capture success.
+
+    The decision to replay (including the validation replay, which
+    EXECUTES any captured collectives) must be collective-consistent:
+    a rank replaying an all-reduce while another rank skips it deadlocks
+    the job. If any rank failed to capture, every rank runs eager — the
+    per-step collective sequence is identical either way, but the extra
+    validation replay is not.
+    """
+    if not (dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1):
+        return captured
+    votes = [None] * dist.get_world_size()
+    dist.all_gather_object(votes, captured)
+    return all(votes)
 
 
 class GraphedStep:
@@ -53,6 +71,7 @@ class GraphedStep:
         self._initialized = True
         if not self.enabled:
             return
+        graph = None
         try:
             stream = torch.cuda.Stream()
             stream.wait_stream(torch.cuda.current_stream())
@@ -65,13 +84,25 @@ class GraphedStep:
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
                 self.step_fn()
-            if self.validate:
-                graph.replay()
-                torch.cuda.synchronize()
-            self.graph = graph
         except Exception as e:  # pragma: no cover - depends on runtime support
             logger.warning(f'hipGraph capture failed ({e!r}); falling back to eager stepping')
+            graph = None
+
+        # all ranks must agree before anything replays (see _capture_unanimous)
+        if not _capture_unanimous(graph is not None):
+            if graph is not None:
+                logger.warning('hipGraph captured here but not on every rank; using eager everywhere')
             self.graph = None
+            return
+
+        if graph is not None and self.validate:
+            try:
+                graph.replay()
+                torch.cuda.synchronize()
+            except Exception as e:  # pragma: no cover
+                logger.warning(f'hipGraph validation replay failed ({e!r}); eager fallback')
+                graph = None
+        self.graph = graph
 
     @property
     def captured(self) -> bool:

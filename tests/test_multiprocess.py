@@ -350,6 +350,17 @@ def _root_helpers(rank, world_size, tmpdir):
         assert gathered == [0, 1]
 
 
+def _graph_capture_vote(rank, world_size, tmpdir):
+    """The hipGraph capture vote: any rank failing capture forces eager
+    EVERYWHERE (a lone validation replay of a captured collective would
+    deadlock the job)."""
+    from dmlcloud_amd.parallel.graphs import _capture_unanimous
+
+    assert _capture_unanimous(True) is True  # all captured -> replay ok
+    assert _capture_unanimous(rank == 0) is False  # rank 1 failed -> all eager
+    assert _capture_unanimous(False) is False
+
+
 def _root_first_ordering(rank, world_size, tmpdir):
     """root_first: rank 0's body completes before any other rank's starts
     (the dataset-download fence)."""
@@ -390,6 +401,7 @@ def _root_first_ordering(rank, world_size, tmpdir):
         '_pipeline_two_ranks',
         '_root_helpers',
         '_root_first_ordering',
+        '_graph_capture_vote',
     ],
 )
 def test_multiprocess(payload, tmp_path):
