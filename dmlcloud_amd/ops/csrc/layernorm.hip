@@ -230,6 +230,14 @@ void layernorm_fwd(at::Tensor x, at::Tensor gamma, at::Tensor beta, at::Tensor y
   const int D = x.size(-1);
   const int64_t R = x.numel() / D;
   TORCH_CHECK(D % 8 == 0 && D <= 8 * kWave * kMaxVec, "D must be a multiple of 8 and <= 2048");
+  TORCH_CHECK(gamma.numel() == D && beta.numel() == D &&
+                  gamma.scalar_type() == at::kBFloat16 && beta.scalar_type() == at::kBFloat16,
+              "gamma/beta must be bf16[D]");
+  TORCH_CHECK(y.numel() == x.numel() && y.scalar_type() == at::kBFloat16 && y.is_contiguous(),
+              "y must be contiguous bf16 like x");
+  TORCH_CHECK(mean.numel() >= R && rstd.numel() >= R && mean.scalar_type() == at::kFloat &&
+                  rstd.scalar_type() == at::kFloat,
+              "mean/rstd must be fp32[R]");
   auto stream = c10::hip::getCurrentHIPStream();
   const int waves_per_block = kBlock / kWave;
   const int blocks = (int)std::min<int64_t>((R + waves_per_block - 1) / waves_per_block, kMaxGrid);
@@ -245,6 +253,11 @@ void layernorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor mean, at::Tensor rstd
                    at::Tensor dbeta) {
   const int D = x.size(-1);
   const int64_t R = x.numel() / D;
+  TORCH_CHECK(dy.numel() == x.numel() && dy.is_contiguous() && dx.numel() == x.numel(),
+              "dy/dx must match x");
+  TORCH_CHECK(mean.numel() >= R && rstd.numel() >= R, "mean/rstd must be fp32[R]");
+  TORCH_CHECK(gamma.numel() == D && dgamma.numel() == D && dbeta.numel() == D,
+              "gamma/dgamma/dbeta must be [D]");
   auto stream = c10::hip::getCurrentHIPStream();
   const int waves_per_block = kBlock / kWave;
   const int blocks =

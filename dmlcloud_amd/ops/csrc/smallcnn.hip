@@ -403,6 +403,12 @@ void conv3x3_relu_pool_fwd(at::Tensor in, at::Tensor w, at::Tensor b, at::Tensor
   const int N = in.size(0), CIN = in.size(1), H = in.size(2), W = in.size(3);
   const int COUT = w.size(0);
   TORCH_CHECK(H % 2 == 0 && W % 2 == 0, "H and W must be even for 2x2 pooling");
+  TORCH_CHECK(w.size(1) == CIN, "weight in-channels must match input");
+  TORCH_CHECK(b.numel() == COUT, "bias must be [COUT]");
+  TORCH_CHECK(out.numel() == (int64_t)N * COUT * (H / 2) * (W / 2) && out.is_contiguous(),
+              "out must be contiguous [N,COUT,H/2,W/2]");
+  TORCH_CHECK(argmax.numel() == out.numel() && argmax.scalar_type() == at::kByte,
+              "argmax must be uint8 like out");
   const PlaneGeom pg = plane_geom(H, W);
   const int lds_bytes = (CIN * pg.cs + COUT * CIN * 9) * (int)sizeof(float);
   TORCH_CHECK(lds_bytes <= kMaxLds, "plane+weights exceed LDS (", lds_bytes, " B)");
@@ -417,6 +423,10 @@ void conv3x3_relu_pool_bwd_data(at::Tensor dpooled, at::Tensor argmax, at::Tenso
                                 at::Tensor w, at::Tensor din) {
   const int N = din.size(0), CIN = din.size(1), H = din.size(2), W = din.size(3);
   const int COUT = w.size(0);
+  TORCH_CHECK(w.size(1) == CIN && din.is_contiguous(), "weight/din geometry mismatch");
+  TORCH_CHECK(dpooled.numel() == (int64_t)N * COUT * (H / 2) * (W / 2) &&
+                  argmax.numel() == dpooled.numel() && pooled.numel() == dpooled.numel(),
+              "dpooled/argmax/pooled must be [N,COUT,H/2,W/2]");
   const PlaneGeom pg = plane_geom(H, W);
   const int lds_bytes = (COUT * pg.cs + COUT * CIN * 9) * (int)sizeof(float);
   TORCH_CHECK(lds_bytes <= kMaxLds, "dconv plane exceeds LDS");
