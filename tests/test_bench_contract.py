@@ -56,6 +56,13 @@ def test_bench_ddp_impl():
     assert result['config']['impl'] == 'ddp'
 
 
+def test_bench_metric_stress_mode():
+    """BASELINE config #5 plumbing: N per-step reducers tracked and
+    reported in the JSON config."""
+    result = _run_bench(['--metric-stress', '8'])
+    assert result['config']['metric_stress'] == 8
+
+
 def _free_port():
     import socket
 

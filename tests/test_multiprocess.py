@@ -439,6 +439,7 @@ def _tracker_fused_anyw(rank, world_size, tmpdir):
         '_flat_matches_ddp_math',
         '_flat_clip_matches_ddp',
         '_flat_overlap_matches_single',
+        '_flat_overlap_no_sync_accumulation',
     ],
 )
 def test_multiprocess_w4(payload, tmp_path):
