@@ -218,8 +218,9 @@ def save_tensor_state(obj, path: Union[str, Path]):
 
     host_buf = torch.empty(total, dtype=torch.uint8, pin_memory=torch.cuda.is_available()) if total else None
 
-    device_tensors = [(i, t) for i, t in enumerate(tensors) if t.is_cuda]
-    cpu_tensors = [(i, t) for i, t in enumerate(tensors) if not t.is_cuda]
+    # zero-element tensors carry no payload (their meta reconstructs them)
+    device_tensors = [(i, t) for i, t in enumerate(tensors) if t.is_cuda and t.numel()]
+    cpu_tensors = [(i, t) for i, t in enumerate(tensors) if not t.is_cuda and t.numel()]
 
     if device_tensors:
         dev = device_tensors[0][1].device
